@@ -1,0 +1,331 @@
+"""Minimal Kubernetes REST client + in-memory fake.
+
+The plugin needs four apiserver verbs (reference uses client-go for the
+same surface, pkg/gpu/nvidia/podmanager.go):
+  - GET  node                              (isolation label check)
+  - PATCH node /status                     (aliyun.com/gpu-count)
+  - LIST pods (fieldSelector, all ns)      (pending-pod fallback path)
+  - PATCH pod (strategic merge)            (ASSIGNED=true handshake)
+
+plus the kubelet read-only endpoint:
+  - GET https://<node>:10250/pods/         (primary pending-pod path,
+    reference: pkg/kubelet/client/client.go:119-134)
+
+``FakeKubeClient`` implements the same surface in memory with
+resourceVersion bumping and injectable 409 conflicts, giving the unit tests
+what the reference never had (no fake clientset anywhere).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import ssl
+import threading
+from typing import Optional
+
+import yaml
+
+SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
+
+
+class KubeError(RuntimeError):
+    def __init__(self, status: int, message: str):
+        super().__init__(f"HTTP {status}: {message}")
+        self.status = status
+
+
+class ConflictError(KubeError):
+    """Optimistic-lock conflict (HTTP 409) on a patch."""
+
+    def __init__(self, message: str = "conflict"):
+        super().__init__(409, message)
+
+
+# --------------------------------------------------------------------------- #
+# Real clients (httpx)
+# --------------------------------------------------------------------------- #
+
+class RestKubeClient:
+    """Apiserver client: in-cluster service account, or a kubeconfig file
+    (env ``KUBECONFIG``), mirroring kubeInit (podmanager.go:29-57)."""
+
+    def __init__(
+        self,
+        base_url: Optional[str] = None,
+        token: Optional[str] = None,
+        verify=None,
+        timeout: float = 10.0,
+    ):
+        import httpx
+
+        if base_url is None:
+            base_url, token, verify = self._auto_config()
+        headers = {"Accept": "application/json"}
+        if token:
+            headers["Authorization"] = f"Bearer {token}"
+        self._client = httpx.Client(
+            base_url=base_url,
+            headers=headers,
+            verify=verify if verify is not None else False,
+            timeout=timeout,
+        )
+
+    @staticmethod
+    def _auto_config():
+        kubeconfig = os.environ.get("KUBECONFIG")
+        if kubeconfig and os.path.exists(kubeconfig):
+            cfg = yaml.safe_load(open(kubeconfig))
+            ctx_name = cfg.get("current-context")
+            ctx = next(
+                c["context"] for c in cfg["contexts"] if c["name"] == ctx_name
+            )
+            cluster = next(
+                c["cluster"]
+                for c in cfg["clusters"]
+                if c["name"] == ctx["cluster"]
+            )
+            user = next(
+                u["user"] for u in cfg["users"] if u["name"] == ctx["user"]
+            )
+            verify: object = cluster.get("certificate-authority", False)
+            if cluster.get("insecure-skip-tls-verify"):
+                verify = False
+            token = user.get("token")
+            if not token and user.get("client-certificate"):
+                sslctx = ssl.create_default_context(
+                    cafile=cluster.get("certificate-authority")
+                )
+                if verify is False:
+                    sslctx.check_hostname = False
+                    sslctx.verify_mode = ssl.CERT_NONE
+                sslctx.load_cert_chain(
+                    user["client-certificate"], user.get("client-key")
+                )
+                verify = sslctx
+            return cluster["server"], token, verify
+        # in-cluster
+        host = os.environ.get("KUBERNETES_SERVICE_HOST", "kubernetes.default.svc")
+        port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
+        token = None
+        token_path = os.path.join(SA_DIR, "token")
+        if os.path.exists(token_path):
+            token = open(token_path).read().strip()
+        ca = os.path.join(SA_DIR, "ca.crt")
+        verify = ca if os.path.exists(ca) else False
+        return f"https://{host}:{port}", token, verify
+
+    # -- verbs ---------------------------------------------------------------
+    def _check(self, resp):
+        if resp.status_code == 409:
+            raise ConflictError(resp.text)
+        if resp.status_code >= 400:
+            raise KubeError(resp.status_code, resp.text)
+        return resp.json()
+
+    def get_node(self, name: str) -> dict:
+        return self._check(self._client.get(f"/api/v1/nodes/{name}"))
+
+    def patch_node_status(self, name: str, patch: dict) -> dict:
+        return self._check(
+            self._client.patch(
+                f"/api/v1/nodes/{name}/status",
+                content=json.dumps(patch),
+                headers={"Content-Type": "application/strategic-merge-patch+json"},
+            )
+        )
+
+    def list_pods(self, field_selector: str = "", namespace: str = "") -> dict:
+        path = (
+            f"/api/v1/namespaces/{namespace}/pods" if namespace else "/api/v1/pods"
+        )
+        params = {"fieldSelector": field_selector} if field_selector else {}
+        return self._check(self._client.get(path, params=params))
+
+    def get_pod(self, namespace: str, name: str) -> dict:
+        return self._check(
+            self._client.get(f"/api/v1/namespaces/{namespace}/pods/{name}")
+        )
+
+    def patch_pod(self, namespace: str, name: str, patch: dict) -> dict:
+        return self._check(
+            self._client.patch(
+                f"/api/v1/namespaces/{namespace}/pods/{name}",
+                content=json.dumps(patch),
+                headers={"Content-Type": "application/strategic-merge-patch+json"},
+            )
+        )
+
+    def close(self) -> None:
+        self._client.close()
+
+
+class KubeletClient:
+    """Read-only kubelet client (GET /pods), bearer-token HTTPS with TLS
+    verification off — matching the reference's forced insecure transport
+    (client.go:75-99; the kubelet's serving cert is rarely CA-signed)."""
+
+    def __init__(
+        self,
+        address: str = "127.0.0.1",
+        port: int = 10250,
+        token: Optional[str] = None,
+        timeout: float = 10.0,
+    ):
+        import httpx
+
+        headers = {"Accept": "application/json"}
+        if token is None:
+            token_path = os.path.join(SA_DIR, "token")
+            if os.path.exists(token_path):
+                token = open(token_path).read().strip()
+        if token:
+            headers["Authorization"] = f"Bearer {token}"
+        self._client = httpx.Client(
+            base_url=f"https://{address}:{port}",
+            headers=headers,
+            verify=False,
+            timeout=timeout,
+        )
+
+    def get_node_running_pods(self) -> dict:
+        resp = self._client.get("/pods/")
+        if resp.status_code >= 400:
+            raise KubeError(resp.status_code, resp.text)
+        return resp.json()
+
+    def close(self) -> None:
+        self._client.close()
+
+
+# --------------------------------------------------------------------------- #
+# Fakes (tests + bench harness)
+# --------------------------------------------------------------------------- #
+
+class FakeKubeClient:
+    """In-memory apiserver: same verbs, resourceVersion bumping, injectable
+    conflicts; doubles as the kubelet /pods source via ``as_kubelet()``."""
+
+    def __init__(self, node_name: str = "node-a"):
+        self.node_name = node_name
+        self._lock = threading.RLock()
+        self._rv = 0
+        self.nodes: dict[str, dict] = {
+            node_name: {
+                "metadata": {"name": node_name, "labels": {}},
+                "status": {"capacity": {}, "allocatable": {}},
+            }
+        }
+        self.pods: dict[tuple, dict] = {}
+        self.fail_next_pod_patches = 0   # inject N consecutive 409s
+        self.patch_count = 0
+        self.list_count = 0
+
+    def _bump(self, obj: dict) -> None:
+        self._rv += 1
+        obj.setdefault("metadata", {})["resourceVersion"] = str(self._rv)
+
+    # -- test helpers --------------------------------------------------------
+    def add_pod(self, pod: dict) -> dict:
+        with self._lock:
+            key = (pod["metadata"].get("namespace", "default"), pod["metadata"]["name"])
+            pod["metadata"].setdefault("namespace", "default")
+            pod["metadata"].setdefault("uid", f"uid-{key[0]}-{key[1]}")
+            self._bump(pod)
+            self.pods[key] = pod
+            return pod
+
+    def delete_pod(self, namespace: str, name: str) -> None:
+        with self._lock:
+            self.pods.pop((namespace, name), None)
+
+    # -- apiserver verbs -----------------------------------------------------
+    def get_node(self, name: str) -> dict:
+        with self._lock:
+            if name not in self.nodes:
+                raise KubeError(404, f"node {name} not found")
+            return json.loads(json.dumps(self.nodes[name]))
+
+    def patch_node_status(self, name: str, patch: dict) -> dict:
+        with self._lock:
+            node = self.nodes[name]
+            for sect in ("capacity", "allocatable"):
+                node["status"].setdefault(sect, {}).update(
+                    patch.get("status", {}).get(sect, {})
+                )
+            self._bump(node)
+            return json.loads(json.dumps(node))
+
+    def list_pods(self, field_selector: str = "", namespace: str = "") -> dict:
+        with self._lock:
+            self.list_count += 1
+            sel = dict(
+                kv.split("=", 1) for kv in field_selector.split(",") if "=" in kv
+            )
+            items = []
+            for pod in self.pods.values():
+                if namespace and pod["metadata"]["namespace"] != namespace:
+                    continue
+                if "spec.nodeName" in sel and (
+                    pod.get("spec", {}).get("nodeName") != sel["spec.nodeName"]
+                ):
+                    continue
+                if "status.phase" in sel and (
+                    pod.get("status", {}).get("phase") != sel["status.phase"]
+                ):
+                    continue
+                items.append(json.loads(json.dumps(pod)))
+            return {"kind": "PodList", "items": items}
+
+    def get_pod(self, namespace: str, name: str) -> dict:
+        with self._lock:
+            key = (namespace, name)
+            if key not in self.pods:
+                raise KubeError(404, f"pod {namespace}/{name} not found")
+            return json.loads(json.dumps(self.pods[key]))
+
+    def patch_pod(self, namespace: str, name: str, patch: dict) -> dict:
+        with self._lock:
+            self.patch_count += 1
+            if self.fail_next_pod_patches > 0:
+                self.fail_next_pod_patches -= 1
+                raise ConflictError()
+            key = (namespace, name)
+            if key not in self.pods:
+                raise KubeError(404, f"pod {namespace}/{name} not found")
+            pod = self.pods[key]
+            anns = patch.get("metadata", {}).get("annotations")
+            if anns:
+                pod["metadata"].setdefault("annotations", {}).update(anns)
+            self._bump(pod)
+            return json.loads(json.dumps(pod))
+
+    def close(self) -> None:
+        pass
+
+    # -- kubelet view --------------------------------------------------------
+    def as_kubelet(self) -> "FakeKubeletClient":
+        return FakeKubeletClient(self)
+
+
+class FakeKubeletClient:
+    def __init__(self, store: FakeKubeClient, fail_times: int = 0):
+        self._store = store
+        self.fail_times = fail_times
+        self.query_count = 0
+
+    def get_node_running_pods(self) -> dict:
+        self.query_count += 1
+        if self.fail_times > 0:
+            self.fail_times -= 1
+            raise KubeError(500, "kubelet unavailable (injected)")
+        with self._store._lock:
+            items = [
+                json.loads(json.dumps(p))
+                for p in self._store.pods.values()
+                if p.get("spec", {}).get("nodeName") == self._store.node_name
+            ]
+        return {"kind": "PodList", "items": items}
+
+    def close(self) -> None:
+        pass

@@ -1,0 +1,87 @@
+"""Wire-contract constants.
+
+The resource names, socket names, and the ``ALIYUN_COM_GPU_MEM_*`` pod
+annotation/env protocol are external contracts shared with the
+gpushare-scheduler-extender and existing workloads; they are kept
+byte-for-byte compatible with the reference
+(reference: pkg/gpu/nvidia/const.go:10-36).  Everything device-side is
+AMD-native.
+"""
+
+# ---------------------------------------------------------------------------
+# Kubelet device-plugin API v1beta1 contract
+# (reference: vendor/k8s.io/kubernetes/pkg/kubelet/apis/deviceplugin/v1beta1/constants.go)
+# ---------------------------------------------------------------------------
+HEALTHY = "Healthy"
+UNHEALTHY = "Unhealthy"
+API_VERSION = "v1beta1"
+DEVICE_PLUGIN_PATH = "/var/lib/kubelet/device-plugins/"
+KUBELET_SOCKET_NAME = "kubelet.sock"
+
+# Our plugin's own unix socket name inside DEVICE_PLUGIN_PATH.
+# (reference: const.go:13 "aliyungpushare.sock"; ours is the AMD daemon)
+SERVER_SOCK_NAME = "amdgpushare.sock"
+
+# Max length of a Device.ID on the wire (api.proto:84).
+MAX_DEVICE_ID_LEN = 63
+
+# ---------------------------------------------------------------------------
+# Extended resources (scheduler contract — unchanged from the reference so the
+# gpushare-scheduler-extender works as-is; reference: const.go:11-12)
+# ---------------------------------------------------------------------------
+RESOURCE_NAME = "aliyun.com/gpu-mem"
+RESOURCE_COUNT = "aliyun.com/gpu-count"
+
+# ---------------------------------------------------------------------------
+# Pod annotation / env protocol (scheduler-extender handshake,
+# reference: const.go:25-31 + podutils.go)
+# ---------------------------------------------------------------------------
+ENV_RESOURCE_INDEX = "ALIYUN_COM_GPU_MEM_IDX"
+ENV_RESOURCE_BY_POD = "ALIYUN_COM_GPU_MEM_POD"
+ENV_RESOURCE_BY_CONTAINER = "ALIYUN_COM_GPU_MEM_CONTAINER"
+ENV_RESOURCE_BY_DEV = "ALIYUN_COM_GPU_MEM_DEV"
+ENV_ASSIGNED_FLAG = "ALIYUN_COM_GPU_MEM_ASSIGNED"
+ENV_RESOURCE_ASSUME_TIME = "ALIYUN_COM_GPU_MEM_ASSUME_TIME"
+ENV_RESOURCE_ASSIGN_TIME = "ALIYUN_COM_GPU_MEM_ASSIGN_TIME"
+
+# Per-container allocation map written by newer scheduler-extender versions
+# (reference: cmd/inspect/nodeinfo.go:244-271): JSON {container: {gpuIdx: mem}}.
+ANN_GPUSHARE_ALLOCATION = "scheduler.framework.gpushare.allocation"
+
+# ---------------------------------------------------------------------------
+# AMD-native container injection (replaces NVIDIA_VISIBLE_DEVICES; on ROCm the
+# env alone does nothing — the /dev/kfd + /dev/dri render nodes must be
+# injected as DeviceSpec entries, which Allocate() does).
+# ---------------------------------------------------------------------------
+ENV_HIP_VISIBLE = "HIP_VISIBLE_DEVICES"
+ENV_ROCR_VISIBLE = "ROCR_VISIBLE_DEVICES"
+
+DEV_KFD = "/dev/kfd"
+DEV_DRI_DIR = "/dev/dri"
+
+# Node label opting a node out of kernel-level isolation (analogue of the
+# reference's cgpu.disable.isolation toggle, podmanager.go:59-72).
+LABEL_DISABLE_ISOLATION = "cgpu.disable.isolation"
+ENV_CGPU_DISABLE = "CGPU_DISABLE"
+
+# ---------------------------------------------------------------------------
+# Failure-path contract: Allocate() failures return a *successful* RPC whose
+# envs poison the pod visibly (reference: allocate.go:24-39).  String format
+# preserved: "no-gpu-has-<N><unit>-to-run".
+# ---------------------------------------------------------------------------
+def poisoned_visible_devices(req_units: int, unit: str) -> str:
+    return f"no-gpu-has-{req_units}{unit}-to-run"
+
+
+# Optimistic-concurrency conflict detection for annotation patches
+# (reference: const.go:15; we match on HTTP 409 primarily, this string as a
+# fallback for proxies that rewrite status codes).
+OPTIMISTIC_LOCK_ERROR_MSG = (
+    "the object has been modified; please apply your changes to the latest "
+    "version and try again"
+)
+
+# Memory units (reference: const.go:34-35).
+GIB = "GiB"
+MIB = "MiB"
+VALID_MEMORY_UNITS = (GIB, MIB)

@@ -1,0 +1,123 @@
+"""Device layer: mock source, fake-device expansion, KFD topology walker."""
+
+import pytest
+
+from gpushare_amd import consts
+from gpushare_amd.device import PhysicalGPU, create_source
+from gpushare_amd.device import fakedev, kfd_topology
+from gpushare_amd.device.mock_source import MockSource
+
+
+def test_mock_spec_parsing():
+    src = MockSource.from_spec("8x288GiB")
+    gpus = src.devices()
+    assert len(gpus) == 8
+    assert gpus[0].memory_bytes == 288 << 30
+    assert gpus[0].mem_units(consts.GIB) == 288
+    assert gpus[0].mem_units(consts.MIB) == 288 * 1024
+
+
+def test_mock_spec_invalid():
+    with pytest.raises(ValueError):
+        MockSource.from_spec("eight-gpus")
+
+
+def test_create_source_env(monkeypatch):
+    monkeypatch.setenv("GPUSHARE_MOCK_SPEC", "1x8GiB")
+    src = create_source()
+    assert len(src.devices()) == 1
+    assert src.devices()[0].mem_units(consts.GIB) == 8
+
+
+def test_fake_id_roundtrip():
+    fid = fakedev.fake_id("amd-00ff", 287)
+    assert fid == "amd-00ff-_-287"
+    assert fakedev.real_id(fid) == "amd-00ff"
+
+
+def test_fake_id_length_cap():
+    with pytest.raises(ValueError):
+        fakedev.fake_id("u" * 62, 1000)
+
+
+def test_expansion_mi355x_node():
+    gpus = MockSource.from_spec("8x288GiB").devices()
+    table = fakedev.FakeDeviceTable.build(gpus, consts.GIB)
+    assert len(table) == 8 * 288
+    assert all(len(i) <= consts.MAX_DEVICE_ID_LEN for i in table.ids)
+    # GPU-major layout and correct back-mapping
+    assert table.gpu_of[table.ids[0]] == 0
+    assert table.gpu_of[table.ids[288]] == 1
+    assert list(table.gpu_fake_indices(2)) == list(range(2 * 288, 3 * 288))
+    assert table.uuid_of[3] == gpus[3].uuid
+    assert table.index_of[gpus[3].uuid] == 3
+
+
+def test_expansion_rejects_zero_memory():
+    gpu = PhysicalGPU(index=0, uuid="z", memory_bytes=0)
+    with pytest.raises(ValueError):
+        fakedev.FakeDeviceTable.build([gpu], consts.GIB)
+
+
+# --------------------------------------------------------------------------- #
+# KFD topology from a fake sysfs tree
+# --------------------------------------------------------------------------- #
+
+def _write(p, text):
+    p.parent.mkdir(parents=True, exist_ok=True)
+    p.write_text(text)
+
+
+@pytest.fixture
+def fake_kfd(tmp_path):
+    """2 GPUs + 1 CPU node; xGMI between the GPUs; renderD129/130."""
+    root = tmp_path / "nodes"
+    # node 0: CPU
+    _write(root / "0" / "gpu_id", "0\n")
+    _write(root / "0" / "properties", "cpu_cores_count 32\nsimd_count 0\n")
+    for n, (gpuid, minor, uid) in enumerate(
+        [(1234, 129, 0xAB01), (5678, 130, 0xAB02)], start=1
+    ):
+        _write(root / str(n) / "gpu_id", f"{gpuid}\n")
+        _write(
+            root / str(n) / "properties",
+            f"simd_count 1024\ngfx_target_version 90500\n"
+            f"drm_render_minor {minor}\nunique_id {uid}\n",
+        )
+        _write(
+            root / str(n) / "mem_banks" / "0" / "properties",
+            f"heap_type 1\nsize_in_bytes {8 << 30}\n",
+        )
+        # xGMI link to the other GPU node, PCIe link to CPU node 0
+        other = 2 if n == 1 else 1
+        _write(
+            root / str(n) / "io_links" / "0" / "properties",
+            f"type 11\nnode_from {n}\nnode_to {other}\nweight 15\n",
+        )
+        _write(
+            root / str(n) / "io_links" / "1" / "properties",
+            f"type 2\nnode_from {n}\nnode_to 0\nweight 20\n",
+        )
+    return str(root)
+
+
+def test_kfd_topology_parse(fake_kfd):
+    nodes = kfd_topology.read_topology(fake_kfd)
+    assert len(nodes) == 3
+    gpus = [n for n in nodes if n.is_gpu]
+    assert len(gpus) == 2
+    assert gpus[0].render_minor == 129
+    assert gpus[0].vram_bytes == 8 << 30
+    assert gpus[0].xgmi_peer_nodes == [2]
+    assert gpus[0].pcie_peer_nodes == [0]
+
+
+def test_kfd_resolve(fake_kfd, tmp_path):
+    topo = kfd_topology.resolve(fake_kfd, drm_root=str(tmp_path / "no-drm"))
+    assert set(topo.keys()) == {1234, 5678}
+    t = topo[1234]
+    assert t.render_path == "/dev/dri/renderD129"
+    assert t.unique_id == 0xAB01
+    assert t.xgmi_peer_gpu_ids == [5678]
+    assert t.vram_bytes == 8 << 30
+    assert t.gfx_target_version == 90500

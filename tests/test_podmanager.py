@@ -1,0 +1,123 @@
+"""PodManager: listing paths, retries, dedup, FIFO sort, cache, patches."""
+
+import pytest
+
+from gpushare_amd import consts
+from gpushare_amd.cluster.kubeclient import FakeKubeClient, FakeKubeletClient
+from gpushare_amd.cluster.podmanager import PodManager
+
+from helpers import make_pod, plain_pod
+
+
+@pytest.fixture
+def kube():
+    return FakeKubeClient(node_name="node-a")
+
+
+def _pm(kube, **kw):
+    kw.setdefault("kubelet_client", kube.as_kubelet())
+    kw.setdefault("cache_ttl", 0.0)  # most tests want fresh lists
+    kw.setdefault("kubelet_retries", 0)
+    kw.setdefault("kubelet_retry_interval", 0.0)
+    kw.setdefault("apiserver_retries", 0)
+    kw.setdefault("apiserver_retry_interval", 0.0)
+    return PodManager(kube, "node-a", **kw)
+
+
+def test_pending_pods_filters_node_and_phase(kube):
+    kube.add_pod(make_pod("on-node", 2))
+    kube.add_pod(make_pod("other-node", 2, node="node-b"))
+    kube.add_pod(plain_pod("running", phase="Running"))
+    pm = _pm(kube)
+    pods = pm.get_pending_pods()
+    assert [p["metadata"]["name"] for p in pods] == ["on-node"]
+
+
+def test_kubelet_fallback_to_apiserver(kube):
+    kube.add_pod(make_pod("p1", 2))
+    kubelet = FakeKubeletClient(kube, fail_times=100)
+    pm = _pm(kube, kubelet_client=kubelet)
+    pods = pm.get_pending_pods()
+    assert len(pods) == 1
+    assert kube.list_count >= 1  # fell back to apiserver LIST
+
+
+def test_kubelet_retry_then_success(kube):
+    kube.add_pod(make_pod("p1", 2))
+    kubelet = FakeKubeletClient(kube, fail_times=2)
+    pm = _pm(kube, kubelet_client=kubelet, kubelet_retries=3)
+    pods = pm.get_pending_pods()
+    assert len(pods) == 1
+    assert kubelet.query_count == 3  # 2 failures + 1 success
+    assert kube.list_count == 0      # never hit apiserver
+
+
+def test_candidate_sort_fifo(kube):
+    kube.add_pod(make_pod("new", 2, assume_time_ns=2000))
+    kube.add_pod(make_pod("old", 2, assume_time_ns=1000))
+    kube.add_pod(make_pod("assigned", 2, assigned="true"))
+    pm = _pm(kube)
+    names = [p["metadata"]["name"] for p in pm.get_candidate_pods()]
+    assert names == ["old", "new"]
+
+
+def test_cache_serves_within_ttl(kube):
+    kube.add_pod(make_pod("p1", 2))
+    kubelet = kube.as_kubelet()
+    pm = _pm(kube, kubelet_client=kubelet, cache_ttl=60.0)
+    pm.get_pending_pods()
+    pm.get_pending_pods()
+    pm.get_pending_pods()
+    assert kubelet.query_count == 1  # served from cache
+    pm.get_pending_pods(force_refresh=True)
+    assert kubelet.query_count == 2
+
+
+def test_mark_assigned_conflict_retry(kube):
+    pod = kube.add_pod(make_pod("p1", 2))
+    kube.fail_next_pod_patches = 1
+    pm = _pm(kube)
+    assert pm.mark_assigned(pod) is True
+    assert kube.patch_count == 2
+    stored = kube.get_pod("default", "p1")
+    assert stored["metadata"]["annotations"][consts.ENV_ASSIGNED_FLAG] == "true"
+
+
+def test_mark_assigned_gives_up_after_retries(kube):
+    pod = kube.add_pod(make_pod("p1", 2))
+    kube.fail_next_pod_patches = 5
+    pm = _pm(kube)
+    assert pm.mark_assigned(pod, retries=1) is False
+
+
+def test_patch_gpu_count(kube):
+    pm = _pm(kube)
+    pm.patch_gpu_count(8)
+    node = kube.get_node("node-a")
+    assert node["status"]["capacity"][consts.RESOURCE_COUNT] == "8"
+    assert node["status"]["allocatable"][consts.RESOURCE_COUNT] == "8"
+    # idempotent second patch short-circuits
+    before = kube._rv
+    pm.patch_gpu_count(8)
+    assert kube._rv == before
+
+
+def test_isolation_disabled_label(kube):
+    pm = _pm(kube)
+    assert pm.isolation_disabled() is False
+    kube.nodes["node-a"]["metadata"]["labels"][consts.LABEL_DISABLE_ISOLATION] = "true"
+    assert pm.isolation_disabled() is True
+
+
+def test_dedup_by_uid(kube):
+    pod = make_pod("dup", 2)
+    kube.add_pod(pod)
+    # simulate kubelet returning the same pod twice
+    class DupKubelet:
+        def get_node_running_pods(self):
+            import json
+            p = json.loads(json.dumps(pod))
+            return {"items": [p, p]}
+
+    pm = _pm(kube, kubelet_client=DupKubelet())
+    assert len(pm.get_pending_pods()) == 1

@@ -1,0 +1,105 @@
+"""Wire-protocol tests: v1beta1 messages + device-list codec.
+
+Oracle: the programmatically-built protobuf classes (which the reference's
+generated Go code and the kubelet both round-trip through)."""
+
+import pytest
+
+from gpushare_amd import consts
+from gpushare_amd.device import fakedev
+from gpushare_amd.deviceplugin import v1beta1 as api
+
+
+def test_device_roundtrip():
+    d = api.Device(ID="amd-x-_-0", health=consts.HEALTHY)
+    d2 = api.Device.FromString(d.SerializeToString())
+    assert d2.ID == "amd-x-_-0" and d2.health == "Healthy"
+
+
+def test_register_request_fields():
+    r = api.RegisterRequest(
+        version="v1beta1",
+        endpoint="amdgpushare.sock",
+        resource_name=consts.RESOURCE_NAME,
+    )
+    r2 = api.RegisterRequest.FromString(r.SerializeToString())
+    assert r2.resource_name == "aliyun.com/gpu-mem"
+    assert r2.version == "v1beta1"
+
+
+def test_allocate_request_grouping():
+    req = api.AllocateRequest()
+    cr = req.container_requests.add()
+    cr.devicesIDs.extend(["a-_-0", "a-_-1"])
+    cr2 = req.container_requests.add()
+    cr2.devicesIDs.append("a-_-2")
+    parsed = api.AllocateRequest.FromString(req.SerializeToString())
+    assert [list(c.devicesIDs) for c in parsed.container_requests] == [
+        ["a-_-0", "a-_-1"],
+        ["a-_-2"],
+    ]
+
+
+def test_container_allocate_response_maps_and_devices():
+    c = api.ContainerAllocateResponse(
+        envs={"HIP_VISIBLE_DEVICES": "0"},
+        devices=[
+            api.DeviceSpec(
+                container_path="/dev/kfd", host_path="/dev/kfd", permissions="rw"
+            )
+        ],
+        annotations={"k": "v"},
+    )
+    c2 = api.ContainerAllocateResponse.FromString(c.SerializeToString())
+    assert c2.envs["HIP_VISIBLE_DEVICES"] == "0"
+    assert c2.devices[0].host_path == "/dev/kfd"
+    assert c2.annotations["k"] == "v"
+
+
+# --------------------------------------------------------------------------- #
+# codec: native vs python vs protobuf byte-identical
+# --------------------------------------------------------------------------- #
+
+IDS = [f"amd-{i:016x}-_-{j}" for i in range(4) for j in range(17)]
+
+
+def _protobuf_encode(ids, unhealthy):
+    return api.ListAndWatchResponse(
+        devices=[
+            api.Device(
+                ID=d,
+                health=consts.UNHEALTHY if k in unhealthy else consts.HEALTHY,
+            )
+            for k, d in enumerate(ids)
+        ]
+    ).SerializeToString()
+
+
+@pytest.mark.parametrize("unhealthy", [set(), {0}, {3, 17, 67}, set(range(68))])
+def test_python_encoder_matches_protobuf(unhealthy):
+    assert fakedev.encode_list_python(IDS, unhealthy) == _protobuf_encode(
+        IDS, unhealthy
+    )
+
+
+@pytest.mark.parametrize("unhealthy", [[], [5], [1, 50]])
+def test_native_codec_matches_protobuf(unhealthy):
+    _devlist = pytest.importorskip("gpushare_amd._devlist")
+    codec = _devlist.DeviceListCodec(IDS)
+    assert codec.encode(unhealthy) == _protobuf_encode(IDS, set(unhealthy))
+
+
+def test_native_codec_rejects_long_ids():
+    _devlist = pytest.importorskip("gpushare_amd._devlist")
+    with pytest.raises(ValueError):
+        _devlist.DeviceListCodec(["x" * 64])
+
+
+def test_codec_scale_2304_devices():
+    """8×MI355X node: 2,304 fake devices must encode fast and correctly."""
+    ids = [f"amd-{i:016x}-_-{j}" for i in range(8) for j in range(288)]
+    codec = fakedev.make_codec(ids)
+    payload = codec.encode([])
+    assert payload == _protobuf_encode(ids, set())
+    parsed = api.ListAndWatchResponse.FromString(payload)
+    assert len(parsed.devices) == 2304
