@@ -90,7 +90,7 @@ bool load_library() {
       "/opt/rocm/lib/libamd_smi.so.26",
   };
   for (const char *c : candidates) {
-    g_lib = dlopen(c, RTLD_LAZY | RTLD_GLOBAL);
+    g_lib = dlopen(c, RTLD_LAZY | RTLD_LOCAL);
     if (g_lib) {
       g_lib_path = c;
       break;
