@@ -142,6 +142,20 @@ class PodManager:
         with self._cache_lock:
             self._cached_pods = None
 
+    def _mark_cached_assigned(self, pod: dict) -> None:
+        """Coherence-only cache update after a successful ASSIGNED patch —
+        keeps steady-state Allocate at zero remote list calls (durable state
+        is the patch itself; the cache refresh would rediscover this)."""
+        uid = podutils.pod_uid(pod)
+        with self._cache_lock:
+            if self._cached_pods is None:
+                return
+            for p in self._cached_pods:
+                if podutils.pod_uid(p) == uid:
+                    p.setdefault("metadata", {}).setdefault("annotations", {})[
+                        consts.ENV_ASSIGNED_FLAG
+                    ] = "true"
+
     def get_candidate_pods(self, force_refresh: bool = False) -> list:
         """Assumed-but-unassigned pods, oldest assume-time first (FIFO
         fairness; reference: getCandidatePods, podmanager.go:215-262)."""
@@ -163,7 +177,7 @@ class PodManager:
         for attempt in range(retries + 1):
             try:
                 self.kube.patch_pod(ns, name, patch)
-                self.invalidate_cache()
+                self._mark_cached_assigned(pod)
                 return True
             except ConflictError:
                 if attempt < retries:

@@ -1,0 +1,122 @@
+"""Extender core: cluster-state sync + the annotation-handshake writer.
+
+Crash-only like the plugin: the ledger is soft state rebuilt from pod
+annotations; the durable record of every placement is the
+``ALIYUN_COM_GPU_MEM_IDX`` / ``ASSUME_TIME`` / ``ASSIGNED`` annotation set
+this class writes (the exact triple the plugin's Allocate consumes,
+reference: podutils.go:78-119).
+"""
+
+from __future__ import annotations
+
+import logging
+import threading
+import time
+from typing import Optional
+
+from .. import consts
+from ..cluster import podutils
+from .binpack import BinpackState
+
+log = logging.getLogger(__name__)
+
+
+class GPUShareExtender:
+    def __init__(self, kube_client, resync_interval: float = 30.0):
+        self.kube = kube_client
+        self.state = BinpackState()
+        self._lock = threading.Lock()
+        self.resync_interval = resync_interval
+        self._last_resync = 0.0
+        self.assumed = 0
+        self.rejected = 0
+
+    # ------------------------------------------------------------------ #
+    # state sync
+    # ------------------------------------------------------------------ #
+    def register_node(self, node: str, per_gpu_units: list[int]) -> None:
+        self.state.set_node(node, per_gpu_units)
+
+    def resync(self, nodes: Optional[list[str]] = None) -> None:
+        """Rebuild the ledger from pod annotations (source of truth)."""
+        nodes = nodes or list(self.state.nodes.keys())
+        pods = self.kube.list_pods().get("items", [])
+        for node in nodes:
+            st = self.state.nodes.get(node)
+            if st is None:
+                continue
+            allocated = [0] * len(st.per_gpu_units)
+            for pod in pods:
+                if pod.get("spec", {}).get("nodeName") != node:
+                    continue
+                if podutils.pod_is_not_running(pod):
+                    continue
+                mem = podutils.gpu_memory_of_pod(pod)
+                if mem <= 0:
+                    continue
+                anns = podutils.annotations(pod)
+                if consts.ENV_RESOURCE_ASSUME_TIME not in anns:
+                    continue
+                idx = podutils.gpu_id_from_annotation(pod)
+                if 0 <= idx < len(allocated):
+                    allocated[idx] += mem
+            self.state.set_node(node, st.per_gpu_units, allocated)
+        self._last_resync = time.monotonic()
+
+    def _maybe_resync(self) -> None:
+        if time.monotonic() - self._last_resync > self.resync_interval:
+            try:
+                self.resync()
+            except Exception as e:  # noqa: BLE001
+                log.warning("extender resync failed: %s", e)
+
+    # ------------------------------------------------------------------ #
+    # scheduling verbs
+    # ------------------------------------------------------------------ #
+    def filter(self, pod: dict, node_names: list[str]) -> list[str]:
+        """Webhook `filter`: nodes with a GPU that fits the pod."""
+        self._maybe_resync()
+        request = podutils.gpu_memory_of_pod(pod)
+        if request <= 0:
+            return node_names
+        return self.state.filter_nodes(request, node_names)
+
+    def assume(self, pod: dict, node: str) -> Optional[int]:
+        """Webhook `bind` body: best-fit a GPU, write the annotation triple.
+        Returns the GPU index or None if the node cannot fit the pod."""
+        request = podutils.gpu_memory_of_pod(pod)
+        if request <= 0:
+            return None
+        idx = self.state.assume(node, request)
+        if idx is None:
+            self.rejected += 1
+            return None
+        ns = podutils.pod_namespace(pod)
+        name = podutils.pod_name(pod)
+        patch = {
+            "metadata": {
+                "annotations": {
+                    consts.ENV_RESOURCE_INDEX: str(idx),
+                    consts.ENV_RESOURCE_ASSUME_TIME: str(time.time_ns()),
+                    consts.ENV_ASSIGNED_FLAG: "false",
+                }
+            }
+        }
+        try:
+            self.kube.patch_pod(ns, name, patch)
+        except Exception as e:  # noqa: BLE001
+            log.warning("assume patch failed for %s/%s: %s", ns, name, e)
+            self.state.release(node, idx, request)
+            return None
+        self.assumed += 1
+        return idx
+
+    def release(self, pod: dict, node: str) -> None:
+        """Informer delete-event path: return the pod's reservation."""
+        request = podutils.gpu_memory_of_pod(pod)
+        idx = podutils.gpu_id_from_annotation(pod)
+        if request > 0 and idx >= 0:
+            self.state.release(node, idx, request)
+
+    def packing(self) -> dict:
+        return self.state.packing()
