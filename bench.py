@@ -60,7 +60,9 @@ def parse_args():
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--pods-per-gpu", type=int, default=4)
-    p.add_argument("--pod-gib", type=int, default=72)
+    p.add_argument("--pod-gib", type=int, default=0,
+               help="0 = derive as advertised_units // pods_per_gpu "
+                    "(287//4=71 on real MI355X: 288 GiB minus reserved)")
     p.add_argument(
         "--mixed",
         action="store_true",
@@ -234,6 +236,8 @@ def main():
     api = RestKubeClient(base_url=info["api_url"])
     ext = ExtenderClient(info["ext_url"])
     plugin_client = DevicePluginClient(info["socket"], consts.RESOURCE_NAME)
+    if args.pod_gib <= 0:
+        args.pod_gib = info["units_per_gpu"] // args.pods_per_gpu
     total_grains = args.gpus * info["units_per_gpu"]
     plugin_client.wait_for_devices(min_count=total_grains, timeout=30)
     grains = sorted(plugin_client.devices)
