@@ -1,0 +1,117 @@
+"""Device-plugin daemon entrypoint (reference: cmd/nvidia/main.go).
+
+Flag parity with the reference (main.go:15-26) minus the dead ``--mps``
+flag (plumbed but never read anywhere in the reference — SURVEY §2 dead
+code; deliberately not carried), plus MI355X-native additions:
+
+  --mock-spec        run against fake GPUs (CI; BASELINE config 1)
+  --deep-probe       active gfx950 canary health probe interval (seconds)
+  --no-inject        don't add /dev/kfd+renderD DeviceSpecs (env-only mode)
+
+Env: NODE_NAME (required — reference crashes at package-import time if
+unset, podmanager.go:52-55; we fail at startup with a clear message),
+KUBECONFIG (optional; in-cluster config otherwise).
+"""
+
+from __future__ import annotations
+
+import argparse
+import logging
+import os
+import sys
+
+from .. import consts
+from ..cluster.kubeclient import KubeletClient, RestKubeClient
+from ..device import create_source
+from ..lifecycle import ManagerOptions, SharedGPUManager
+
+
+def parse_args(argv=None):
+    p = argparse.ArgumentParser(
+        prog="amdgpushare-device-plugin",
+        description="MI355X GPU-sharing device plugin (aliyun.com/gpu-mem)",
+    )
+    p.add_argument("--health-check", action="store_true",
+                   help="watch amdsmi events (thermal/reset/ECC) for device health")
+    p.add_argument("--deep-probe", type=float, default=0.0, metavar="SECONDS",
+                   help="also run the gfx950 canary kernels (MFMA+VRAM) on "
+                        "every GPU at this interval; 0 disables")
+    p.add_argument("--memory-unit", default=consts.GIB,
+                   choices=list(consts.VALID_MEMORY_UNITS),
+                   help="granularity of the gpu-mem resource (reference "
+                        "main.go:67-78 validation)")
+    p.add_argument("--query-kubelet", action="store_true",
+                   help="list pending pods from the kubelet read-only API "
+                        "instead of the apiserver")
+    p.add_argument("--kubelet-address", default="127.0.0.1")
+    p.add_argument("--kubelet-port", type=int, default=10250)
+    p.add_argument("--client-cert", default="")
+    p.add_argument("--client-key", default="")
+    p.add_argument("--token", default="")
+    p.add_argument("--timeout", type=int, default=10,
+                   help="kubelet client timeout (seconds)")
+    p.add_argument("--socket-dir", default=consts.DEVICE_PLUGIN_PATH)
+    p.add_argument("--cache-ttl", type=float, default=0.2,
+                   help="pending-pod cache TTL (seconds); 0 = list per Allocate "
+                        "(reference behavior)")
+    p.add_argument("--mock-spec", default=None,
+                   help="e.g. 1x8GiB: serve fake devices (no GPU needed)")
+    p.add_argument("--no-inject", action="store_true",
+                   help="skip /dev/kfd+/dev/dri DeviceSpec injection")
+    p.add_argument("-v", "--verbose", action="count", default=0)
+    return p.parse_args(argv)
+
+
+def main(argv=None) -> int:
+    args = parse_args(argv)
+    logging.basicConfig(
+        level=logging.DEBUG if args.verbose else logging.INFO,
+        format="%(asctime)s %(levelname)s %(name)s: %(message)s",
+        stream=sys.stderr,
+    )
+    log = logging.getLogger("daemon")
+
+    node_name = os.environ.get("NODE_NAME")
+    if not node_name:
+        log.error("NODE_NAME env is required (set via downward API in the "
+                  "DaemonSet, deploy/device-plugin-ds.yaml)")
+        return 2
+
+    source = create_source(args.mock_spec)
+    kube = RestKubeClient()
+    kubelet = None
+    if args.query_kubelet:
+        kubelet = KubeletClient(
+            address=args.kubelet_address,
+            port=args.kubelet_port,
+            token=args.token or None,
+            timeout=args.timeout,
+        )
+
+    mgr = SharedGPUManager(
+        source,
+        kube,
+        node_name,
+        kubelet_client=kubelet,
+        options=ManagerOptions(
+            memory_unit=args.memory_unit,
+            query_kubelet=args.query_kubelet,
+            health_check=args.health_check,
+            deep_probe_interval=args.deep_probe,
+            socket_dir=args.socket_dir,
+            cache_ttl=args.cache_ttl,
+            inject_devices=not args.no_inject,
+        ),
+    )
+    mgr.install_signal_handlers()
+    log.info("starting gpushare device plugin on node %s", node_name)
+    try:
+        mgr.run()
+    except RuntimeError as e:
+        log.error("fatal: %s", e)
+        return 1
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

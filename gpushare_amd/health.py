@@ -1,0 +1,99 @@
+"""Health monitoring — passive amdsmi events + optional active canary probes.
+
+Reference analogue: the ``healthcheck()`` goroutine bridging watchXIDs to
+the plugin's health channel (server.go:203-221, nvidia.go:100-152), with two
+MI355X-native upgrades:
+
+- a physical-GPU event flips **all** of that GPU's fake devices (plugin
+  handles the fan-out; reference flipped one);
+- an optional *active* probe launches the gfx950 canary kernels
+  (``_canary.probe``: MFMA bit-exactness + VRAM pattern walk) on every GPU
+  at a configurable interval — catching the shared-GPU failure mode a
+  passive watcher cannot see: a device that still enumerates but no longer
+  computes correctly.  Recovery flips the device back to Healthy (reference
+  FIXME server.go:180: Unhealthy was terminal).
+"""
+
+from __future__ import annotations
+
+import logging
+import threading
+
+log = logging.getLogger(__name__)
+
+
+class HealthMonitor:
+    def __init__(
+        self,
+        source,
+        plugin,
+        deep_probe_interval: float = 0.0,   # 0 = passive only
+        probe_vram_mb: int = 32,
+    ):
+        self.source = source
+        self.plugin = plugin
+        self.deep_probe_interval = deep_probe_interval
+        self.probe_vram_mb = probe_vram_mb
+        self._stop = threading.Event()
+        self._threads: list[threading.Thread] = []
+        self._probe_failed: set[int] = set()
+
+    # ------------------------------------------------------------------ #
+    def start(self) -> None:
+        self._stop.clear()
+        t = threading.Thread(target=self._watch_passive, name="health-passive",
+                             daemon=True)
+        t.start()
+        self._threads = [t]
+        if self.deep_probe_interval > 0:
+            t2 = threading.Thread(target=self._probe_loop, name="health-probe",
+                                  daemon=True)
+            t2.start()
+            self._threads.append(t2)
+
+    def stop(self) -> None:
+        self._stop.set()
+        for t in self._threads:
+            t.join(timeout=5)
+        self._threads = []
+
+    # ------------------------------------------------------------------ #
+    def _watch_passive(self) -> None:
+        try:
+            for ev in self.source.watch_health(self._stop):
+                log.warning(
+                    "health event: gpu=%s healthy=%s kind=%s %s",
+                    ev.gpu_index, ev.healthy, ev.kind, ev.message,
+                )
+                self.plugin.set_gpu_health(ev.gpu_index, ev.healthy)
+        except Exception as e:  # noqa: BLE001
+            log.error("passive health watcher died: %s", e)
+
+    def _probe_loop(self) -> None:
+        try:
+            import gpushare_amd._canary as canary
+        except ImportError as e:
+            log.error("deep probe requested but _canary not built: %s", e)
+            return
+        while not self._stop.wait(self.deep_probe_interval):
+            for gpu in self.plugin.gpus:
+                if self._stop.is_set():
+                    return
+                try:
+                    result = canary.probe(
+                        gpu.index, vram_probe_mb=self.probe_vram_mb,
+                        bandwidth=False,
+                    )
+                    ok = bool(result.get("ok"))
+                except RuntimeError as e:
+                    log.error("canary probe GPU %d errored: %s", gpu.index, e)
+                    ok = False
+                    result = {"error": str(e)}
+                if not ok and gpu.index not in self._probe_failed:
+                    log.error("deep probe FAILED on GPU %d: %s", gpu.index, result)
+                    self._probe_failed.add(gpu.index)
+                    self.plugin.set_gpu_health(gpu.index, False)
+                elif ok and gpu.index in self._probe_failed:
+                    log.warning("deep probe recovered on GPU %d", gpu.index)
+                    self._probe_failed.discard(gpu.index)
+                    self.plugin.set_gpu_health(gpu.index, True)
