@@ -1,0 +1,26 @@
+# MI355X gpushare device plugin
+PY ?= python3
+
+.PHONY: build test test-gpu bench lint image clean
+
+build:
+	$(PY) -m gpushare_amd.native.build
+
+test: build
+	$(PY) -m pytest tests/ -q -m "not gpu"
+
+test-gpu: build
+	$(PY) -m pytest tests/ -q -m gpu
+
+bench: build
+	$(PY) bench.py --gpus 1 --steps 10 --warmup 2
+
+lint:
+	$(PY) -m compileall -q gpushare_amd tests bench.py __graft_entry__.py
+
+image:
+	docker build -t gpushare/amd-device-plugin:latest .
+
+clean:
+	rm -f gpushare_amd/*.so
+	find . -name __pycache__ -type d -exec rm -rf {} +

@@ -74,6 +74,9 @@ class _Handler(BaseHTTPRequestHandler):
             return
         rest = parts[2:]
 
+        if rest == ["nodes"] and method == "GET":
+            self._send(200, s.list_nodes())
+            return
         if rest[:1] == ["nodes"] and len(rest) >= 2:
             name = rest[1]
             if method == "GET" and len(rest) == 2:

@@ -126,6 +126,9 @@ class RestKubeClient:
     def get_node(self, name: str) -> dict:
         return self._check(self._client.get(f"/api/v1/nodes/{name}"))
 
+    def list_nodes(self) -> dict:
+        return self._check(self._client.get("/api/v1/nodes"))
+
     def patch_node_status(self, name: str, patch: dict) -> dict:
         return self._check(
             self._client.patch(
@@ -245,6 +248,13 @@ class FakeKubeClient:
             if name not in self.nodes:
                 raise KubeError(404, f"node {name} not found")
             return json.loads(json.dumps(self.nodes[name]))
+
+    def list_nodes(self) -> dict:
+        with self._lock:
+            return {
+                "kind": "NodeList",
+                "items": [json.loads(json.dumps(n)) for n in self.nodes.values()],
+            }
 
     def patch_node_status(self, name: str, patch: dict) -> dict:
         with self._lock:

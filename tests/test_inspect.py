@@ -1,0 +1,116 @@
+"""Inspect CLI: data model, annotation precedence, summary/details output."""
+
+import io
+import json
+
+from gpushare_amd import consts
+from gpushare_amd.cli import inspect as insp
+from gpushare_amd.cluster.kubeclient import FakeKubeClient
+
+from helpers import make_pod
+
+
+def _cluster():
+    kube = FakeKubeClient("node-a")
+    kube.nodes["node-a"]["status"] = {
+        "capacity": {consts.RESOURCE_COUNT: "2", consts.RESOURCE_NAME: "16"},
+        "allocatable": {consts.RESOURCE_COUNT: "2", consts.RESOURCE_NAME: "16"},
+        "addresses": [{"type": "InternalIP", "address": "10.0.0.5"}],
+    }
+    kube.nodes["plain-node"] = {
+        "metadata": {"name": "plain-node", "labels": {}},
+        "status": {"capacity": {}, "allocatable": {}},
+    }
+    return kube
+
+
+def test_get_allocation_precedence():
+    # allocation-map annotation wins
+    pod = make_pod(
+        "p", 6, gpu_idx=0,
+        extra_annotations={
+            consts.ANN_GPUSHARE_ALLOCATION: json.dumps({"c0": {"1": 4}, "c1": {"1": 2}})
+        },
+    )
+    assert insp.get_allocation(pod) == {1: 6}
+    # fallback to IDX
+    assert insp.get_allocation(make_pod("p", 3, gpu_idx=1)) == {1: 3}
+    # no annotation -> pending bucket
+    pod = make_pod("p", 3)
+    del pod["metadata"]["annotations"][consts.ENV_RESOURCE_INDEX]
+    assert insp.get_allocation(pod) == {insp.PENDING: 3}
+
+
+def test_summary_output():
+    kube = _cluster()
+    kube.add_pod(make_pod("a", 3, gpu_idx=0, assigned="true", phase="Running"))
+    kube.add_pod(make_pod("b", 2, gpu_idx=0, assigned="true", phase="Running"))
+    kube.add_pod(make_pod("c", 4, gpu_idx=1, assigned="true", phase="Running"))
+    # terminal pod must not count
+    kube.add_pod(make_pod("dead", 8, gpu_idx=1, phase="Failed"))
+    out = io.StringIO()
+    assert insp.main([], kube=kube, out=out) == 0
+    text = out.getvalue()
+    assert "GPU0(Allocated/Total)" in text
+    assert "5/8" in text       # GPU0: 3+2 of 8
+    assert "4/8" in text       # GPU1
+    assert "9/16 (56%)" in text
+    assert "plain-node" not in text   # not a sharing node
+
+
+def test_details_output():
+    kube = _cluster()
+    kube.add_pod(make_pod("a", 3, gpu_idx=0, assigned="true", phase="Running"))
+    pending = make_pod("pend", 2, phase="Pending")
+    del pending["metadata"]["annotations"][consts.ENV_RESOURCE_INDEX]
+    kube.add_pod(pending)
+    out = io.StringIO()
+    assert insp.main(["-d"], kube=kube, out=out) == 0
+    text = out.getvalue()
+    assert "NAME:       node-a" in text
+    assert "IPADDRESS:  10.0.0.5" in text
+    assert "Pending(Allocated)" in text
+    assert "pend" in text
+    assert "Allocated : 5 (31%)" in text
+
+
+def test_node_filter_and_missing():
+    kube = _cluster()
+    out = io.StringIO()
+    assert insp.main(["no-such-node"], kube=kube, out=out) == 1
+    assert "No shared-GPU nodes" in out.getvalue()
+
+
+def test_unit_inference_mi355x_gib():
+    """287 GiB per GPU must NOT be misread as MiB (reference threshold 100
+    would; ours is 1024)."""
+    kube = FakeKubeClient("node-a")
+    kube.nodes["node-a"]["status"] = {
+        "capacity": {},
+        "allocatable": {
+            consts.RESOURCE_COUNT: "8",
+            consts.RESOURCE_NAME: str(287 * 8),
+        },
+    }
+    infos = insp.build_node_infos(kube)
+    assert insp.infer_memory_unit(infos) == consts.GIB
+    # true-MiB node
+    kube.nodes["node-a"]["status"]["allocatable"][consts.RESOURCE_NAME] = str(
+        287 * 1024 * 8
+    )
+    infos = insp.build_node_infos(kube)
+    assert insp.infer_memory_unit(infos) == consts.MIB
+
+
+def test_inspect_over_http_apiserver():
+    from gpushare_amd.cluster.fakeapiserver import FakeApiServer
+
+    kube = _cluster()
+    kube.add_pod(make_pod("h", 4, gpu_idx=1, assigned="true", phase="Running"))
+    api = FakeApiServer(store=kube).start()
+    try:
+        out = io.StringIO()
+        assert insp.main(["--api-url", api.url], kube=None, out=out) == 0
+        assert "4/8" in out.getvalue()
+    finally:
+        api.stop()
