@@ -31,7 +31,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
-import statistics
+import subprocess
 import sys
 import tempfile
 import time
@@ -43,13 +43,11 @@ import torch.distributed as dist
 
 from gpushare_amd import consts
 from gpushare_amd.allocator import Allocator
-from gpushare_amd.cluster.fakeapiserver import FakeApiServer
 from gpushare_amd.cluster.kubeclient import RestKubeClient
 from gpushare_amd.cluster.podmanager import PodManager
 from gpushare_amd.deviceplugin.stubkubelet import DevicePluginClient
 from gpushare_amd.deviceplugin.server import GPUSharePlugin
-from gpushare_amd.extender.core import GPUShareExtender
-from gpushare_amd.extender.server import ExtenderClient, ExtenderServer
+from gpushare_amd.extender.server import ExtenderClient
 
 NODE = "bench-node"
 
@@ -63,6 +61,9 @@ def parse_args():
     p.add_argument("--pod-gib", type=int, default=0,
                help="0 = derive as advertised_units // pods_per_gpu "
                     "(287//4=71 on real MI355X: 288 GiB minus reserved)")
+    p.add_argument("--rate", type=float, default=0.0,
+                   help="cap total pod churn at this many pods/s across all "
+                        "ranks (BASELINE config 5: 10); 0 = unthrottled")
     p.add_argument(
         "--mixed",
         action="store_true",
@@ -119,6 +120,20 @@ def make_pod_spec(name: str, mem: int) -> dict:
 MIXED_SIZES = [8, 16, 32, 64, 128]
 
 
+class RateLimiter:
+    """Paces one rank's pod submissions at rate/world pods/s."""
+
+    def __init__(self, per_rank_rate: float):
+        self.interval = 1.0 / per_rank_rate
+        self.next_t = time.perf_counter()
+
+    def wait(self):
+        now = time.perf_counter()
+        if now < self.next_t:
+            time.sleep(self.next_t - now)
+        self.next_t = max(self.next_t + self.interval, now - 5 * self.interval)
+
+
 def run_wave(
     rank: int,
     step: int,
@@ -129,6 +144,7 @@ def run_wave(
     lat_allocate: list,
     lat_extender: list,
     grains: list[str],
+    rate_limiter=None,
 ) -> tuple[int, int, float]:
     """One churn wave for this rank.  Returns (allocated, failed, peak_packing)."""
     n_pods = args.pods_per_gpu
@@ -145,6 +161,8 @@ def run_wave(
     allocated = failed = 0
     # create + bind + allocate
     for name, mem in pods:
+        if rate_limiter is not None:
+            rate_limiter.wait()
         api._client.post(
             "/api/v1/namespaces/default/pods", content=json.dumps(make_pod_spec(name, mem))
         )
@@ -187,32 +205,44 @@ def main():
 
     sync = []
     if rank == 0:
-        # --- rank 0: apiserver + extender + plugin -------------------------
+        # --- rank 0: apiserver proc + extender proc + in-proc plugin ------
+        # apiserver and extender run as their own processes (as in a real
+        # cluster) so the plugin's GIL is not shared with them.
         gpus, source_kind = get_gpus(args.gpus, args.mock)
-        api_server = FakeApiServer(port=0)
-        api_server.store.node_name = NODE
-        api_server.store.nodes = {
-            NODE: {
-                "metadata": {"name": NODE, "labels": {}},
-                "status": {"capacity": {}, "allocatable": {}},
-            }
-        }
-        api_server.start()
+        repo = os.path.dirname(os.path.abspath(__file__))
+        api_proc = subprocess.Popen(
+            [sys.executable, "-m", "gpushare_amd.cluster.fakeapiserver",
+             "--node", NODE],
+            stdout=subprocess.PIPE, text=True, cwd=repo,
+        )
+        api_url = api_proc.stdout.readline().split()[1]
 
-        kube0 = RestKubeClient(base_url=api_server.url)
-        extender = GPUShareExtender(kube0, resync_interval=3600.0)
-        extender.register_node(NODE, [g.mem_units(consts.GIB) for g in gpus])
-        ext_server = ExtenderServer(extender).start()
+        kube0 = RestKubeClient(base_url=api_url)
+        units = [g.mem_units(consts.GIB) for g in gpus]
+        # advertise node resources (what kubelet would do from ListAndWatch +
+        # the plugin's patch_gpu_count) so the extender daemon discovers them
+        kube0.patch_node_status(NODE, {"status": {
+            "capacity": {consts.RESOURCE_COUNT: str(len(gpus)),
+                         consts.RESOURCE_NAME: str(sum(units))},
+            "allocatable": {consts.RESOURCE_COUNT: str(len(gpus)),
+                            consts.RESOURCE_NAME: str(sum(units))},
+        }})
+        ext_proc = subprocess.Popen(
+            [sys.executable, "-m", "gpushare_amd.extender",
+             "--api-url", api_url, "--port", "0",
+             "--resync-interval", "3600"],
+            stdout=subprocess.PIPE, text=True, cwd=repo,
+        )
+        ext_url = ext_proc.stdout.readline().split()[1]
 
         pm = PodManager(
-            RestKubeClient(base_url=api_server.url),
+            RestKubeClient(base_url=api_url),
             NODE,
             kubelet_client=None,
             query_kubelet=False,
             cache_ttl=0.05,
             apiserver_retries=0,
         )
-        pm.patch_gpu_count(len(gpus))
         sockdir = tempfile.mkdtemp(prefix="gpushare-bench-")
         plugin = GPUSharePlugin(
             gpus, Allocator(gpus, pm), socket_dir=sockdir
@@ -220,8 +250,8 @@ def main():
         plugin.start()
         sync = [
             {
-                "api_url": api_server.url,
-                "ext_url": ext_server.url,
+                "api_url": api_url,
+                "ext_url": ext_url,
                 "socket": plugin.socket_path,
                 "source": source_kind,
                 "units_per_gpu": gpus[0].mem_units(consts.GIB),
@@ -244,11 +274,14 @@ def main():
 
     lat_allocate: list[float] = []
     lat_extender: list[float] = []
+    limiter = RateLimiter(args.rate / world) if args.rate > 0 else None
 
     # --- warmup -----------------------------------------------------------
     for step in range(args.warmup):
+        if distributed:
+            dist.barrier()
         run_wave(rank, -1 - step, api, ext, plugin_client, args,
-                 lat_allocate, lat_extender, grains)
+                 lat_allocate, lat_extender, grains, limiter)
     lat_allocate.clear()
     lat_extender.clear()
     if rank == 0:
@@ -263,8 +296,11 @@ def main():
     allocated = failed = 0
     packing_samples = []
     for step in range(args.steps):
+        if distributed:
+            dist.barrier()  # align waves: capacity invariant needs all ranks'
+                            # deletes from wave N-1 done before wave N creates
         a, f, peak = run_wave(rank, step, api, ext, plugin_client, args,
-                              lat_allocate, lat_extender, grains)
+                              lat_allocate, lat_extender, grains, limiter)
         allocated += a
         failed += f
         packing_samples.append(peak)
@@ -338,6 +374,7 @@ def main():
                     else f"{args.pods_per_gpu}/GPU × {args.pod_gib}GiB binpack"
                 ),
                 "device_source": info["source"],
+                "rate_cap_pods_per_s": args.rate or None,
                 "pods_allocated": total_alloc,
                 "pods_failed": total_failed,
                 "allocate_p50_ms": round(pct(all_lat, 0.50), 3),
@@ -360,8 +397,10 @@ def main():
         dist.destroy_process_group()
     if rank == 0:
         plugin.stop()
-        ext_server.stop()
-        api_server.stop()
+        ext_proc.terminate()
+        api_proc.terminate()
+        ext_proc.wait(timeout=5)
+        api_proc.wait(timeout=5)
 
 
 if __name__ == "__main__":

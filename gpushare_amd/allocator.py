@@ -96,10 +96,17 @@ class Allocator:
         self.disable_isolation = disable_isolation
         self.inject_devices = inject_devices
         self.stats = AllocateStats()
-        # Serialize matching+patch so two concurrent Allocates can't claim the
-        # same assumed pod (reference holds one RWMutex, server.go:34).  The
-        # remote list itself is cached, so the critical section is short.
+        # Matching runs under a short in-memory critical section; the
+        # ASSIGNED patch happens OUTSIDE it.  A matched pod is "claimed"
+        # (uid -> expiry) so concurrent Allocates skip it; claims are
+        # process-local soft state — if we crash before the patch lands the
+        # pod stays assumed and the kubelet retries, so crash-only semantics
+        # are preserved.  (The reference instead holds one RWMutex across
+        # its 1-2 remote lists + patch, server.go:34 / allocate.go:59-62 —
+        # that serialization is its p99 floor.)
         self._lock = threading.Lock()
+        self._claims: dict[str, float] = {}
+        self.claim_ttl = 30.0
 
     # ------------------------------------------------------------------ #
     def allocate(self, request) -> "api.AllocateResponse":
@@ -110,55 +117,75 @@ class Allocator:
         list_t = patch_t = 0.0
         ok = False
         try:
-            with self._lock:
-                tl = time.perf_counter()
-                pod = self._match_pod(req_units)
-                list_t = time.perf_counter() - tl
-                if pod is not None:
-                    gpu = self._gpu_for_pod(pod)
-                    if gpu is None:
-                        return self._err_response(request, req_units)
-                    resp = self._build_response(request, req_units, gpu)
-                    tp = time.perf_counter()
-                    if not self.pods.mark_assigned(pod):
-                        patch_t = time.perf_counter() - tp
-                        return self._err_response(request, req_units)
-                    patch_t = time.perf_counter() - tp
-                    ok = True
-                    return resp
-                if len(self.gpus) == 1:
-                    # single-GPU fast path (allocate.go:151-178)
-                    gpu = next(iter(self.gpus.values()))
-                    ok = True
-                    return self._build_response(request, req_units, gpu)
-                log.warning(
-                    "invalid allocation request: %d %s cannot be matched to "
-                    "an assumed pod",
-                    req_units,
-                    self.unit,
-                )
-                return self._err_response(request, req_units)
+            tl = time.perf_counter()
+            pod = self._match_and_claim(req_units)
+            list_t = time.perf_counter() - tl
+            if pod is not None:
+                uid = podutils.pod_uid(pod)
+                gpu = self._gpu_for_pod(pod)
+                if gpu is None:
+                    self._unclaim(uid)
+                    return self._err_response(request, req_units)
+                resp = self._build_response(request, req_units, gpu)
+                tp = time.perf_counter()
+                patched = self.pods.mark_assigned(pod)
+                patch_t = time.perf_counter() - tp
+                if not patched:
+                    self._unclaim(uid)
+                    return self._err_response(request, req_units)
+                ok = True
+                return resp
+            if len(self.gpus) == 1:
+                # single-GPU fast path (allocate.go:151-178)
+                gpu = next(iter(self.gpus.values()))
+                ok = True
+                return self._build_response(request, req_units, gpu)
+            log.warning(
+                "invalid allocation request: %d %s cannot be matched to "
+                "an assumed pod",
+                req_units,
+                self.unit,
+            )
+            return self._err_response(request, req_units)
         finally:
             self.stats.record(time.perf_counter() - t0, list_t, patch_t, ok)
 
     # ------------------------------------------------------------------ #
-    def _match_pod(self, req_units: int) -> Optional[dict]:
-        """Oldest assumed pod with exact total-memory equality; one forced
-        cache revalidation on miss (the TTL cache may trail the extender)."""
-        for force in (False, True):
+    def _match_and_claim(self, req_units: int) -> Optional[dict]:
+        """Oldest unclaimed assumed pod with exact total-memory equality;
+        forced cache revalidation on miss (the TTL cache may trail the
+        extender), retried briefly: a concurrent Allocate may claim a pod
+        from our snapshot while its own (newer) pod is not yet in it — the
+        next snapshot covers that bind.  The listing itself runs outside
+        the claim lock."""
+        backoffs = (0.0, 0.0, 0.01, 0.04)
+        for attempt, force in enumerate((False, True, True, True)):
+            if backoffs[attempt]:
+                time.sleep(backoffs[attempt])
             try:
                 candidates = self.pods.get_candidate_pods(force_refresh=force)
             except Exception as e:  # listing failed entirely
                 log.warning("failed to list candidate pods: %s", e)
                 return None
-            for pod in candidates:
-                if podutils.gpu_memory_of_pod(pod) == req_units:
+            with self._lock:
+                now = time.monotonic()
+                for pod in candidates:
+                    if podutils.gpu_memory_of_pod(pod) != req_units:
+                        continue
+                    uid = podutils.pod_uid(pod)
+                    if self._claims.get(uid, 0.0) > now:
+                        continue  # being handled by a concurrent Allocate
+                    self._claims[uid] = now + self.claim_ttl
+                    if len(self._claims) > 10_000:
+                        self._claims = {
+                            u: t for u, t in self._claims.items() if t > now
+                        }
                     return pod
-            if candidates and not force:
-                # a fresh list won't add pods the cache already had unless
-                # the extender just assumed one — still worth one refresh
-                continue
         return None
+
+    def _unclaim(self, uid: str) -> None:
+        with self._lock:
+            self._claims.pop(uid, None)
 
     def _gpu_for_pod(self, pod: dict) -> Optional[PhysicalGPU]:
         idx = podutils.gpu_id_from_annotation(pod)

@@ -148,3 +148,34 @@ class FakeApiServer:
     def stop(self) -> None:
         self._httpd.shutdown()
         self._httpd.server_close()
+
+
+def main(argv=None) -> int:
+    """Standalone fake apiserver: `python -m gpushare_amd.cluster.fakeapiserver`.
+
+    Used by bench.py so the apiserver is its own process (as in a real
+    cluster) instead of sharing the plugin's GIL."""
+    import argparse
+    import sys
+    import time
+
+    p = argparse.ArgumentParser(prog="gpushare-fake-apiserver")
+    p.add_argument("--port", type=int, default=0)
+    p.add_argument("--node", default="bench-node")
+    args = p.parse_args(argv)
+
+    store = FakeKubeClient(node_name=args.node)
+    server = FakeApiServer(store=store, port=args.port).start()
+    print(f"READY {server.url}", flush=True)
+    try:
+        while True:
+            time.sleep(3600)
+    except KeyboardInterrupt:
+        server.stop()
+    return 0
+
+
+if __name__ == "__main__":
+    import sys
+
+    sys.exit(main())

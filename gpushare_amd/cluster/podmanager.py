@@ -59,9 +59,10 @@ class PodManager:
         self.kubelet_retry_interval = kubelet_retry_interval
         self.apiserver_retries = apiserver_retries
         self.apiserver_retry_interval = apiserver_retry_interval
-        self._cache_lock = threading.Lock()
+        self._cache_lock = threading.Condition()
         self._cached_pods: Optional[list] = None
-        self._cache_time = 0.0
+        self._cache_time = 0.0          # when the cached list STARTED
+        self._refreshing = False
 
     # ------------------------------------------------------------------ #
     # listing
@@ -101,42 +102,63 @@ class PodManager:
 
     def get_pending_pods(self, force_refresh: bool = False) -> list:
         """Pending pods on this node, deduped by UID (reference:
-        getPendingPodsInNode, podmanager.go:162-212), TTL-cached."""
-        now = time.monotonic()
-        with self._cache_lock:
-            if (
-                not force_refresh
-                and self._cached_pods is not None
-                and now - self._cache_time < self.cache_ttl
-            ):
-                return self._cached_pods
+        getPendingPodsInNode, podmanager.go:162-212), TTL-cached.
 
-        if self.query_kubelet:
-            try:
-                pods = self._list_via_kubelet()
-            except (KubeError, OSError):
-                log.warning(
-                    "kubelet /pods failed after retries; falling back to apiserver"
-                )
+        Remote lists are single-flighted: N concurrent callers (N parallel
+        Allocates missing the cache at once) trigger ONE remote list; a
+        forced caller is satisfied only by a list that *started after it
+        asked* (a list already in flight may predate the caller's own
+        bind)."""
+        entered = time.monotonic()
+        with self._cache_lock:
+            while True:
+                if self._cached_pods is not None:
+                    if force_refresh:
+                        if self._cache_time >= entered:
+                            return self._cached_pods
+                    elif entered - self._cache_time < self.cache_ttl:
+                        return self._cached_pods
+                if not self._refreshing:
+                    self._refreshing = True
+                    break
+                self._cache_cond_wait()
+
+        started = time.monotonic()
+        result = None
+        try:
+            if self.query_kubelet:
+                try:
+                    pods = self._list_via_kubelet()
+                except (KubeError, OSError):
+                    log.warning(
+                        "kubelet /pods failed after retries; "
+                        "falling back to apiserver"
+                    )
+                    pods = self._list_via_apiserver()
+            else:
                 pods = self._list_via_apiserver()
-        else:
-            pods = self._list_via_apiserver()
 
-        seen: set = set()
-        result = []
-        for pod in pods:
-            if pod.get("spec", {}).get("nodeName") != self.node_name:
-                continue
-            uid = podutils.pod_uid(pod)
-            if uid in seen:
-                continue
-            seen.add(uid)
-            result.append(pod)
+            seen: set = set()
+            result = []
+            for pod in pods:
+                if pod.get("spec", {}).get("nodeName") != self.node_name:
+                    continue
+                uid = podutils.pod_uid(pod)
+                if uid in seen:
+                    continue
+                seen.add(uid)
+                result.append(pod)
+            return result
+        finally:
+            with self._cache_lock:
+                self._refreshing = False
+                if result is not None:
+                    self._cached_pods = result
+                    self._cache_time = started
+                self._cache_lock.notify_all()
 
-        with self._cache_lock:
-            self._cached_pods = result
-            self._cache_time = time.monotonic()
-        return result
+    def _cache_cond_wait(self) -> None:
+        self._cache_lock.wait(timeout=5.0)
 
     def invalidate_cache(self) -> None:
         with self._cache_lock:

@@ -41,7 +41,7 @@ class GPUSharePlugin:
         resource_name: str = consts.RESOURCE_NAME,
         socket_dir: str = consts.DEVICE_PLUGIN_PATH,
         socket_name: str = consts.SERVER_SOCK_NAME,
-        max_workers: int = 8,
+        max_workers: int = 64,  # each ListAndWatch stream pins a worker thread
     ):
         self.gpus = gpus
         self.allocator = allocator

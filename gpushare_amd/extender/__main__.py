@@ -49,6 +49,7 @@ def main(argv=None) -> int:
     server = ExtenderServer(extender, port=args.port)
     server._thread.daemon = False
     server.start()
+    print(f"READY {server.url}", flush=True)
     try:
         while True:
             time.sleep(args.resync_interval)
