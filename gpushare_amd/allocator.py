@@ -32,7 +32,7 @@ import threading
 import time
 from typing import Optional
 
-from . import consts
+from . import consts, metrics
 from .cluster import podutils
 from .device import PhysicalGPU
 from .deviceplugin import v1beta1 as api
@@ -64,6 +64,7 @@ class AllocateStats:
             self.latencies.append(total)
             if len(self.latencies) > 100_000:
                 del self.latencies[: len(self.latencies) // 2]
+        metrics.observe_allocate(total, list_t, patch_t, ok)
 
     def snapshot(self) -> dict:
         with self.lock:
@@ -89,12 +90,14 @@ class Allocator:
         unit: str = consts.GIB,
         disable_isolation: bool = False,
         inject_devices: bool = True,
+        event_recorder=None,
     ):
         self.gpus = {g.index: g for g in gpus}
         self.pods = pod_manager
         self.unit = unit
         self.disable_isolation = disable_isolation
         self.inject_devices = inject_devices
+        self.events = event_recorder
         self.stats = AllocateStats()
         # Matching runs under a short in-memory critical section; the
         # ASSIGNED patch happens OUTSIDE it.  A matched pod is "claimed"
@@ -233,6 +236,11 @@ class Allocator:
         """Poisoned-env failure response (reference: buildErrResponse,
         allocate.go:24-39 — same string format, AMD env names)."""
         poison = consts.poisoned_visible_devices(req_units, self.unit)
+        if self.events is not None:
+            self.events.emit(
+                "GPUShareAllocateFailed",
+                f"no assumed pod matches a request for {req_units}{self.unit}",
+            )
         responses = api.AllocateResponse()
         for cr in request.container_requests:
             c = responses.container_responses.add()

@@ -58,6 +58,8 @@ def parse_args(argv=None):
                    help="e.g. 1x8GiB: serve fake devices (no GPU needed)")
     p.add_argument("--no-inject", action="store_true",
                    help="skip /dev/kfd+/dev/dri DeviceSpec injection")
+    p.add_argument("--metrics-port", type=int, default=0,
+                   help="serve Prometheus /metrics on this port (0 = off)")
     p.add_argument("-v", "--verbose", action="count", default=0)
     return p.parse_args(argv)
 
@@ -76,6 +78,11 @@ def main(argv=None) -> int:
         log.error("NODE_NAME env is required (set via downward API in the "
                   "DaemonSet, deploy/device-plugin-ds.yaml)")
         return 2
+
+    if args.metrics_port:
+        from .. import metrics
+
+        metrics.serve(args.metrics_port)
 
     source = create_source(args.mock_spec)
     kube = RestKubeClient()

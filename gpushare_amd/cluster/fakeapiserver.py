@@ -88,6 +88,10 @@ class _Handler(BaseHTTPRequestHandler):
         if rest == ["pods"] and method == "GET":
             self._send(200, s.list_pods(q.get("fieldSelector", [""])[0]))
             return
+        if rest[:1] == ["namespaces"] and len(rest) == 3 and rest[2] == "events":
+            if method == "POST":
+                self._send(201, s.create_event(rest[1], self._body()))
+                return
         if rest[:1] == ["namespaces"] and len(rest) >= 3 and rest[2] == "pods":
             ns = rest[1]
             if len(rest) == 3 and method == "POST":

@@ -159,6 +159,15 @@ class RestKubeClient:
             )
         )
 
+    def create_event(self, namespace: str, event: dict) -> dict:
+        return self._check(
+            self._client.post(
+                f"/api/v1/namespaces/{namespace}/events",
+                content=json.dumps(event),
+                headers={"Content-Type": "application/json"},
+            )
+        )
+
     def close(self) -> None:
         self._client.close()
 
@@ -220,6 +229,7 @@ class FakeKubeClient:
             }
         }
         self.pods: dict[tuple, dict] = {}
+        self.events: list[dict] = []
         self.fail_next_pod_patches = 0   # inject N consecutive 409s
         self.patch_count = 0
         self.list_count = 0
@@ -309,6 +319,12 @@ class FakeKubeClient:
                 pod["metadata"].setdefault("annotations", {}).update(anns)
             self._bump(pod)
             return json.loads(json.dumps(pod))
+
+    def create_event(self, namespace: str, event: dict) -> dict:
+        with self._lock:
+            event.setdefault("metadata", {})["namespace"] = namespace
+            self.events.append(event)
+            return json.loads(json.dumps(event))
 
     def close(self) -> None:
         pass

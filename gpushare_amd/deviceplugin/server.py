@@ -24,7 +24,7 @@ from typing import Optional
 
 import grpc
 
-from .. import consts
+from .. import consts, metrics
 from ..device import PhysicalGPU
 from ..device.fakedev import FakeDeviceTable, make_codec
 from . import v1beta1 as api
@@ -54,6 +54,7 @@ class GPUSharePlugin:
 
         self.table = FakeDeviceTable.build(gpus, unit)
         self._codec = make_codec(self.table.ids)
+        metrics.observe_inventory(len(self.table))
 
         # health state: plugin GPU index -> healthy?
         self._health_lock = threading.Condition()
@@ -84,6 +85,10 @@ class GPUSharePlugin:
             if changed:
                 self._health_version += 1
                 self._health_lock.notify_all()
+                metrics.observe_health_event(
+                    "recovered" if healthy else "unhealthy",
+                    len(self._unhealthy_gpus),
+                )
 
     def _unhealthy_fake_indices(self) -> list[int]:
         out: list[int] = []
@@ -111,6 +116,7 @@ class GPUSharePlugin:
             len(self.table),
             len(payload),
         )
+        metrics.observe_law_send()
         yield payload
         while not self._stop_event.is_set() and context.is_active():
             with self._health_lock:
@@ -125,6 +131,7 @@ class GPUSharePlugin:
                 "ListAndWatch: health change -> resend (%d unhealthy grains)",
                 unhealthy,
             )
+            metrics.observe_law_send()
             yield payload
 
     def _allocate(self, request, context) -> "api.AllocateResponse":

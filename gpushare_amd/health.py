@@ -29,11 +29,13 @@ class HealthMonitor:
         plugin,
         deep_probe_interval: float = 0.0,   # 0 = passive only
         probe_vram_mb: int = 32,
+        event_recorder=None,
     ):
         self.source = source
         self.plugin = plugin
         self.deep_probe_interval = deep_probe_interval
         self.probe_vram_mb = probe_vram_mb
+        self.events = event_recorder
         self._stop = threading.Event()
         self._threads: list[threading.Thread] = []
         self._probe_failed: set[int] = set()
@@ -66,6 +68,12 @@ class HealthMonitor:
                     ev.gpu_index, ev.healthy, ev.kind, ev.message,
                 )
                 self.plugin.set_gpu_health(ev.gpu_index, ev.healthy)
+                if self.events is not None:
+                    self.events.emit(
+                        "GPUHealthy" if ev.healthy else "GPUUnhealthy",
+                        f"GPU {ev.gpu_index}: {ev.kind} {ev.message}".strip(),
+                        etype="Normal" if ev.healthy else "Warning",
+                    )
         except Exception as e:  # noqa: BLE001
             log.error("passive health watcher died: %s", e)
 
@@ -93,6 +101,11 @@ class HealthMonitor:
                     log.error("deep probe FAILED on GPU %d: %s", gpu.index, result)
                     self._probe_failed.add(gpu.index)
                     self.plugin.set_gpu_health(gpu.index, False)
+                    if self.events is not None:
+                        self.events.emit(
+                            "GPUCanaryFailed",
+                            f"GPU {gpu.index} deep probe failed: {result}",
+                        )
                 elif ok and gpu.index in self._probe_failed:
                     log.warning("deep probe recovered on GPU %d", gpu.index)
                     self._probe_failed.discard(gpu.index)

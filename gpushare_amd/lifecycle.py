@@ -26,6 +26,7 @@ from dataclasses import dataclass
 
 from . import consts
 from .allocator import Allocator
+from .cluster.events import EventRecorder
 from .cluster.podmanager import PodManager
 from .coredump import coredump
 from .deviceplugin.server import GPUSharePlugin
@@ -85,6 +86,7 @@ class SharedGPUManager:
             unit=self.opt.memory_unit,
             disable_isolation=pm.isolation_disabled(),
             inject_devices=self.opt.inject_devices,
+            event_recorder=EventRecorder(self.kube, self.node_name),
         )
         return GPUSharePlugin(
             gpus,
@@ -101,6 +103,7 @@ class SharedGPUManager:
                 self.source,
                 self.plugin,
                 deep_probe_interval=self.opt.deep_probe_interval,
+                event_recorder=EventRecorder(self.kube, self.node_name),
             )
             self.health.start()
 
