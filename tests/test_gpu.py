@@ -165,3 +165,66 @@ def test_e2e_allocate_and_tenant_workloads(source, tmp_path):
     finally:
         plugin.stop()
         kubelet.stop()
+
+
+def test_manager_e2e_on_gpu(source, tmp_path):
+    """Full lifecycle manager with the real device source on an MI355X."""
+    import threading
+
+    from gpushare_amd import consts
+    from gpushare_amd.cluster.kubeclient import FakeKubeClient
+    from gpushare_amd.deviceplugin.stubkubelet import StubKubelet
+    from gpushare_amd.lifecycle import ManagerOptions, SharedGPUManager
+
+    sockdir = str(tmp_path / "dp")
+    kube = FakeKubeClient(node_name="gpu-node")
+    mgr = SharedGPUManager(
+        source, kube, "gpu-node",
+        options=ManagerOptions(socket_dir=sockdir, cache_ttl=0.0,
+                               health_check=True),
+    )
+    kubelet = StubKubelet(sockdir)
+    os.makedirs(sockdir, exist_ok=True)
+    kubelet.start()
+    t = threading.Thread(target=mgr.run, daemon=True)
+    t.start()
+    try:
+        client = kubelet.wait_for_plugin(consts.RESOURCE_NAME, timeout=30)
+        n = len(client.wait_for_devices(min_count=1, timeout=30))
+        total = sum(g.mem_units("GiB") for g in source.devices())
+        assert n == total
+        node = kube.get_node("gpu-node")
+        assert node["status"]["capacity"][consts.RESOURCE_COUNT] == str(
+            len(source.devices())
+        )
+    finally:
+        mgr.shutdown()
+        t.join(timeout=10)
+        kubelet.stop()
+
+
+def test_health_monitor_deep_probe_on_gpu(source, tmp_path):
+    """Active canary probing against the real GPU: healthy stays healthy."""
+    import time as _time
+
+    from gpushare_amd.allocator import Allocator
+    from gpushare_amd.cluster.kubeclient import FakeKubeClient
+    from gpushare_amd.cluster.podmanager import PodManager
+    from gpushare_amd.deviceplugin.server import GPUSharePlugin
+    from gpushare_amd.health import HealthMonitor
+
+    gpus = source.devices()
+    kube = FakeKubeClient("gpu-node")
+    pm = PodManager(kube, "gpu-node", kubelet_client=kube.as_kubelet(),
+                    cache_ttl=0.0)
+    plugin = GPUSharePlugin(gpus, Allocator(gpus, pm),
+                            socket_dir=str(tmp_path))
+    mon = HealthMonitor(source, plugin, deep_probe_interval=0.5,
+                        probe_vram_mb=16)
+    mon.start()
+    try:
+        _time.sleep(2.5)  # several probe cycles
+        assert plugin._unhealthy_gpus == set()
+        assert mon._probe_failed == set()
+    finally:
+        mon.stop()
