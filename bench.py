@@ -145,6 +145,7 @@ def run_wave(
     lat_extender: list,
     grains: list[str],
     rate_limiter=None,
+    barrier=None,
 ) -> tuple[int, int, float]:
     """One churn wave for this rank.  Returns (allocated, failed, peak_packing)."""
     n_pods = args.pods_per_gpu
@@ -182,6 +183,9 @@ def run_wave(
         else:
             allocated += 1
     peak = ext.packing()["packing_pct"]
+    if barrier is not None:
+        barrier()  # all ranks fully allocated -> peak sample is honest
+        peak = max(peak, ext.packing()["packing_pct"])
     # delete + release
     for name, mem in pods:
         try:
@@ -281,7 +285,8 @@ def main():
         if distributed:
             dist.barrier()
         run_wave(rank, -1 - step, api, ext, plugin_client, args,
-                 lat_allocate, lat_extender, grains, limiter)
+                 lat_allocate, lat_extender, grains, limiter,
+                 dist.barrier if distributed else None)
     lat_allocate.clear()
     lat_extender.clear()
     if rank == 0:
@@ -300,7 +305,8 @@ def main():
             dist.barrier()  # align waves: capacity invariant needs all ranks'
                             # deletes from wave N-1 done before wave N creates
         a, f, peak = run_wave(rank, step, api, ext, plugin_client, args,
-                              lat_allocate, lat_extender, grains, limiter)
+                              lat_allocate, lat_extender, grains, limiter,
+                              dist.barrier if distributed else None)
         allocated += a
         failed += f
         packing_samples.append(peak)
