@@ -164,10 +164,13 @@ def run_wave(
     for name, mem in pods:
         if rate_limiter is not None:
             rate_limiter.wait()
+        spec = make_pod_spec(name, mem)
         api._client.post(
-            "/api/v1/namespaces/default/pods", content=json.dumps(make_pod_spec(name, mem))
+            "/api/v1/namespaces/default/pods", content=json.dumps(spec)
         )
         t0 = time.perf_counter()
+        # real scheduler flow: filter (carries the full pod) then bind
+        ext.filter(spec, [NODE])
         err = ext.bind("default", name, NODE)
         lat_extender.append(time.perf_counter() - t0)
         if err:

@@ -39,6 +39,13 @@ class _Handler(BaseHTTPRequestHandler):
         pass
 
     # -- helpers -------------------------------------------------------------
+    def _send_raw(self, code: int, body: bytes) -> None:
+        self.send_response(code)
+        self.send_header("Content-Type", "application/json")
+        self.send_header("Content-Length", str(len(body)))
+        self.end_headers()
+        self.wfile.write(body)
+
     def _send(self, code: int, obj) -> None:
         body = json.dumps(obj).encode()
         self.send_response(code)
@@ -67,7 +74,7 @@ class _Handler(BaseHTTPRequestHandler):
 
         if method == "GET" and parts == ["pods"]:
             # kubelet read-only view
-            self._send(200, s.as_kubelet().get_node_running_pods())
+            self._send_raw(200, s.kubelet_pods_raw())
             return
         if parts[:2] != ["api", "v1"]:
             self._send(404, {"message": "not found"})
@@ -86,7 +93,7 @@ class _Handler(BaseHTTPRequestHandler):
                 self._send(200, s.patch_node_status(name, self._body()))
                 return
         if rest == ["pods"] and method == "GET":
-            self._send(200, s.list_pods(q.get("fieldSelector", [""])[0]))
+            self._send_raw(200, s.list_pods_raw(q.get("fieldSelector", [""])[0]))
             return
         if rest[:1] == ["namespaces"] and len(rest) == 3 and rest[2] == "events":
             if method == "POST":
@@ -100,12 +107,12 @@ class _Handler(BaseHTTPRequestHandler):
                 self._send(201, s.add_pod(pod))
                 return
             if len(rest) == 3 and method == "GET":
-                self._send(200, s.list_pods(namespace=ns))
+                self._send_raw(200, s.list_pods_raw(namespace=ns))
                 return
             if len(rest) == 4:
                 name = rest[3]
                 if method == "GET":
-                    self._send(200, s.get_pod(ns, name))
+                    self._send_raw(200, s.get_pod_raw(ns, name))
                     return
                 if method == "PATCH":
                     self._send(200, s.patch_pod(ns, name, self._body()))

@@ -229,6 +229,7 @@ class FakeKubeClient:
             }
         }
         self.pods: dict[tuple, dict] = {}
+        self._pod_bytes: dict[tuple, bytes] = {}   # key -> encoded JSON
         self.events: list[dict] = []
         self.fail_next_pod_patches = 0   # inject N consecutive 409s
         self.patch_count = 0
@@ -238,6 +239,9 @@ class FakeKubeClient:
         self._rv += 1
         obj.setdefault("metadata", {})["resourceVersion"] = str(self._rv)
 
+    def _reencode(self, key: tuple) -> None:
+        self._pod_bytes[key] = json.dumps(self.pods[key]).encode()
+
     # -- test helpers --------------------------------------------------------
     def add_pod(self, pod: dict) -> dict:
         with self._lock:
@@ -246,11 +250,13 @@ class FakeKubeClient:
             pod["metadata"].setdefault("uid", f"uid-{key[0]}-{key[1]}")
             self._bump(pod)
             self.pods[key] = pod
+            self._reencode(key)
             return pod
 
     def delete_pod(self, namespace: str, name: str) -> None:
         with self._lock:
             self.pods.pop((namespace, name), None)
+            self._pod_bytes.pop((namespace, name), None)
 
     # -- apiserver verbs -----------------------------------------------------
     def get_node(self, name: str) -> dict:
@@ -318,7 +324,39 @@ class FakeKubeClient:
             if anns:
                 pod["metadata"].setdefault("annotations", {}).update(anns)
             self._bump(pod)
+            self._reencode(key)
             return json.loads(json.dumps(pod))
+
+    # -- raw-bytes accessors (fakeapiserver hot path: the per-pod JSON is
+    # encoded once per mutation, a LIST response is a join of cached bytes) --
+    def list_pods_raw(self, field_selector: str = "", namespace: str = "") -> bytes:
+        sel = dict(
+            kv.split("=", 1) for kv in field_selector.split(",") if "=" in kv
+        )
+        want_node = sel.get("spec.nodeName")
+        want_phase = sel.get("status.phase")
+        with self._lock:
+            self.list_count += 1
+            chunks = []
+            for key, pod in self.pods.items():
+                if namespace and key[0] != namespace:
+                    continue
+                if want_node and pod.get("spec", {}).get("nodeName") != want_node:
+                    continue
+                if want_phase and pod.get("status", {}).get("phase") != want_phase:
+                    continue
+                chunks.append(self._pod_bytes[key])
+            return b'{"kind":"PodList","items":[' + b",".join(chunks) + b"]}"
+
+    def kubelet_pods_raw(self) -> bytes:
+        return self.list_pods_raw(field_selector=f"spec.nodeName={self.node_name}")
+
+    def get_pod_raw(self, namespace: str, name: str) -> bytes:
+        with self._lock:
+            key = (namespace, name)
+            if key not in self.pods:
+                raise KubeError(404, f"pod {namespace}/{name} not found")
+            return self._pod_bytes[key]
 
     def create_event(self, namespace: str, event: dict) -> dict:
         with self._lock:

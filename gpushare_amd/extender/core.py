@@ -30,6 +30,10 @@ class GPUShareExtender:
         self._last_resync = 0.0
         self.assumed = 0
         self.rejected = 0
+        # pods seen at filter time, reused at bind time (the scheduler always
+        # filters before binding) — saves the bind-path GET to the apiserver
+        self._pod_cache: dict[tuple, tuple] = {}
+        self.pod_cache_ttl = 30.0
 
     # ------------------------------------------------------------------ #
     # state sync
@@ -76,6 +80,15 @@ class GPUShareExtender:
     def filter(self, pod: dict, node_names: list[str]) -> list[str]:
         """Webhook `filter`: nodes with a GPU that fits the pod."""
         self._maybe_resync()
+        name = podutils.pod_name(pod)
+        if name:
+            key = (podutils.pod_namespace(pod), name)
+            self._pod_cache[key] = (pod, time.monotonic())
+            if len(self._pod_cache) > 10_000:
+                cutoff = time.monotonic() - self.pod_cache_ttl
+                self._pod_cache = {
+                    k: v for k, v in self._pod_cache.items() if v[1] > cutoff
+                }
         request = podutils.gpu_memory_of_pod(pod)
         if request <= 0:
             return node_names
@@ -117,6 +130,16 @@ class GPUShareExtender:
         idx = podutils.gpu_id_from_annotation(pod)
         if request > 0 and idx >= 0:
             self.state.release(node, idx, request)
+
+    def cached_pod(self, namespace: str, name: str):
+        """Pod captured at filter time, if still fresh (one-shot)."""
+        entry = self._pod_cache.pop((namespace, name), None)
+        if entry is None:
+            return None
+        pod, t = entry
+        if time.monotonic() - t > self.pod_cache_ttl:
+            return None
+        return pod
 
     def packing(self) -> dict:
         return self.state.packing()

@@ -64,7 +64,9 @@ class _Handler(BaseHTTPRequestHandler):
                 ns = body.get("PodNamespace", "default")
                 name = body.get("PodName", "")
                 node = body.get("Node", "")
-                pod = self.extender.kube.get_pod(ns, name)
+                pod = self.extender.cached_pod(ns, name)
+                if pod is None:
+                    pod = self.extender.kube.get_pod(ns, name)
                 idx = self.extender.assume(pod, node)
                 if idx is None:
                     self._send(
