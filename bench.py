@@ -160,6 +160,7 @@ def run_wave(
         pods.append((f"bench-r{rank}-s{step}-p{i}", mem))
 
     allocated = failed = 0
+    placed: dict = {}
     # create + bind + allocate
     for name, mem in pods:
         if rate_limiter is not None:
@@ -185,18 +186,20 @@ def run_wave(
             failed += 1
         else:
             allocated += 1
+            placed[name] = (mem, envs[consts.ENV_RESOURCE_INDEX])
     peak = ext.packing()["packing_pct"]
     if barrier is not None:
         barrier()  # all ranks fully allocated -> peak sample is honest
         peak = max(peak, ext.packing()["packing_pct"])
-    # delete + release
+    # delete + release (release stub built locally — the informer's delete
+    # event carries the pod object, no extra apiserver GET needed)
     for name, mem in pods:
-        try:
-            pod = api.get_pod("default", name)
-        except Exception:
-            continue
         api._client.delete(f"/api/v1/namespaces/default/pods/{name}")
-        ext.release(pod, NODE)
+        if name in placed:
+            mem_placed, idx = placed[name]
+            stub = make_pod_spec(name, mem_placed)
+            stub["metadata"]["annotations"] = {consts.ENV_RESOURCE_INDEX: idx}
+            ext.release(stub, NODE)
     return allocated, failed, peak
 
 

@@ -150,14 +150,20 @@ class RestKubeClient:
             self._client.get(f"/api/v1/namespaces/{namespace}/pods/{name}")
         )
 
-    def patch_pod(self, namespace: str, name: str, patch: dict) -> dict:
-        return self._check(
-            self._client.patch(
-                f"/api/v1/namespaces/{namespace}/pods/{name}",
-                content=json.dumps(patch),
-                headers={"Content-Type": "application/strategic-merge-patch+json"},
-            )
+    def patch_pod(self, namespace: str, name: str, patch: dict,
+                  parse: bool = True):
+        resp = self._client.patch(
+            f"/api/v1/namespaces/{namespace}/pods/{name}",
+            content=json.dumps(patch),
+            headers={"Content-Type": "application/strategic-merge-patch+json"},
         )
+        if parse:
+            return self._check(resp)
+        if resp.status_code == 409:
+            raise ConflictError(resp.text)
+        if resp.status_code >= 400:
+            raise KubeError(resp.status_code, resp.text)
+        return None
 
     def create_event(self, namespace: str, event: dict) -> dict:
         return self._check(
@@ -310,7 +316,8 @@ class FakeKubeClient:
                 raise KubeError(404, f"pod {namespace}/{name} not found")
             return json.loads(json.dumps(self.pods[key]))
 
-    def patch_pod(self, namespace: str, name: str, patch: dict) -> dict:
+    def patch_pod(self, namespace: str, name: str, patch: dict,
+                  parse: bool = True):
         with self._lock:
             self.patch_count += 1
             if self.fail_next_pod_patches > 0:
@@ -325,7 +332,7 @@ class FakeKubeClient:
                 pod["metadata"].setdefault("annotations", {}).update(anns)
             self._bump(pod)
             self._reencode(key)
-            return json.loads(json.dumps(pod))
+            return json.loads(json.dumps(pod)) if parse else None
 
     # -- raw-bytes accessors (fakeapiserver hot path: the per-pod JSON is
     # encoded once per mutation, a LIST response is a join of cached bytes) --

@@ -198,7 +198,7 @@ class PodManager:
         patch = podutils.assigned_patch()
         for attempt in range(retries + 1):
             try:
-                self.kube.patch_pod(ns, name, patch)
+                self.kube.patch_pod(ns, name, patch, parse=False)
                 self._mark_cached_assigned(pod)
                 return True
             except ConflictError:

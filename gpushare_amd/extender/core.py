@@ -116,7 +116,7 @@ class GPUShareExtender:
             }
         }
         try:
-            self.kube.patch_pod(ns, name, patch)
+            self.kube.patch_pod(ns, name, patch, parse=False)
         except Exception as e:  # noqa: BLE001
             log.warning("assume patch failed for %s/%s: %s", ns, name, e)
             self.state.release(node, idx, request)
