@@ -1,0 +1,200 @@
+"""Scale axes and concurrency edge cases (SURVEY §7 hard parts)."""
+
+import json
+import threading
+import time
+
+import pytest
+
+from gpushare_amd import consts
+from gpushare_amd.allocator import Allocator
+from gpushare_amd.cluster.kubeclient import FakeKubeClient
+from gpushare_amd.cluster.podmanager import PodManager
+from gpushare_amd.device import fakedev
+from gpushare_amd.device.mock_source import MockSource
+from gpushare_amd.deviceplugin import v1beta1 as api
+
+from helpers import make_pod
+from test_allocator import _request
+
+
+# --------------------------------------------------------------------------- #
+# grain-count scale (hard part 2)
+# --------------------------------------------------------------------------- #
+
+def test_mib_mode_expansion_fits_id_budget():
+    """MiB granularity on MI355X: 294,912 grains/GPU must stay within the
+    63-char Device.ID budget."""
+    gpus = MockSource.from_spec("1x288GiB").devices()
+    table = fakedev.FakeDeviceTable.build(gpus, consts.MIB)
+    assert len(table) == 288 * 1024
+    assert max(len(i) for i in table.ids) <= consts.MAX_DEVICE_ID_LEN
+
+
+def test_node_payload_encode_and_kubelet_ingest_speed():
+    """8×288 grains: steady-state encode must be trivial (cached buffer) and
+    kubelet-side parse comfortably sub-100ms."""
+    ids = [f"amd-{i:016x}-_-{j}" for i in range(8) for j in range(288)]
+    codec = fakedev.make_codec(ids)
+    t0 = time.perf_counter()
+    for _ in range(100):
+        payload = codec.encode([])
+    encode_us = (time.perf_counter() - t0) / 100 * 1e6
+    t0 = time.perf_counter()
+    parsed = api.ListAndWatchResponse.FromString(payload)
+    parse_ms = (time.perf_counter() - t0) * 1e3
+    assert len(parsed.devices) == 2304
+    assert encode_us < 5000, f"steady-state encode {encode_us:.0f}us"
+    assert parse_ms < 100, f"kubelet-side parse {parse_ms:.1f}ms"
+
+
+def test_health_flip_reencode_speed():
+    ids = [f"amd-{i:016x}-_-{j}" for i in range(8) for j in range(288)]
+    codec = fakedev.make_codec(ids)
+    unhealthy = list(range(288))  # one whole GPU
+    t0 = time.perf_counter()
+    payload = codec.encode(unhealthy)
+    ms = (time.perf_counter() - t0) * 1e3
+    parsed = api.ListAndWatchResponse.FromString(payload)
+    assert sum(1 for d in parsed.devices if d.health == "Unhealthy") == 288
+    assert ms < 50, f"health-flip re-encode took {ms:.1f}ms"
+
+
+# --------------------------------------------------------------------------- #
+# allocate disambiguation race (hard part 1)
+# --------------------------------------------------------------------------- #
+
+def _alloc(kube, spec="8x288GiB"):
+    pm = PodManager(
+        kube, "node-a", kubelet_client=kube.as_kubelet(), cache_ttl=0.0,
+        kubelet_retries=0, kubelet_retry_interval=0.0, apiserver_retries=0,
+    )
+    gpus = MockSource.from_spec(spec).devices()
+    return Allocator(gpus, pm)
+
+
+def test_concurrent_allocates_never_double_assign():
+    """16 same-size pods, 16 concurrent Allocates: every pod assigned exactly
+    once, no poisoned responses."""
+    kube = FakeKubeClient("node-a")
+    alloc = _alloc(kube)
+    for i in range(16):
+        kube.add_pod(make_pod(f"p{i}", 4, gpu_idx=i % 8, assume_time_ns=i))
+    results = []
+
+    def run():
+        resp = alloc.allocate(_request([4]))
+        results.append(resp.container_responses[0].envs[consts.ENV_RESOURCE_INDEX])
+
+    threads = [threading.Thread(target=run) for _ in range(16)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert "-1" not in results
+    assigned = [
+        p for p in kube.pods.values()
+        if p["metadata"]["annotations"][consts.ENV_ASSIGNED_FLAG] == "true"
+    ]
+    assert len(assigned) == 16
+
+
+def test_claim_released_on_patch_failure():
+    """A failed ASSIGNED patch must unclaim the pod so a retry can take it."""
+    kube = FakeKubeClient("node-a")
+    alloc = _alloc(kube)
+    kube.add_pod(make_pod("p0", 4, gpu_idx=0))
+    kube.fail_next_pod_patches = 2  # consume the first allocate's 1-retry budget
+    resp = alloc.allocate(_request([4]))
+    assert resp.container_responses[0].envs[consts.ENV_RESOURCE_INDEX] == "-1"
+    # kubelet retries the Allocate; claim must not block the pod
+    resp = alloc.allocate(_request([4]))
+    assert resp.container_responses[0].envs[consts.ENV_RESOURCE_INDEX] == "0"
+
+
+def test_claim_expires():
+    kube = FakeKubeClient("node-a")
+    alloc = _alloc(kube)
+    alloc.claim_ttl = 0.05
+    pod = kube.add_pod(make_pod("p0", 4, gpu_idx=0))
+    with alloc._lock:
+        alloc._claims[pod["metadata"]["uid"]] = time.monotonic() + 0.05
+    time.sleep(0.08)
+    resp = alloc.allocate(_request([4]))
+    assert resp.container_responses[0].envs[consts.ENV_RESOURCE_INDEX] == "0"
+
+
+# --------------------------------------------------------------------------- #
+# single-flight listing
+# --------------------------------------------------------------------------- #
+
+def test_single_flight_collapses_concurrent_lists():
+    kube = FakeKubeClient("node-a")
+    kube.add_pod(make_pod("p0", 4, gpu_idx=0))
+
+    class SlowKubelet:
+        def __init__(self, inner):
+            self.inner = inner
+            self.calls = 0
+
+        def get_node_running_pods(self):
+            self.calls += 1
+            time.sleep(0.05)
+            return self.inner.get_node_running_pods()
+
+    slow = SlowKubelet(kube.as_kubelet())
+    pm = PodManager(kube, "node-a", kubelet_client=slow, cache_ttl=10.0,
+                    kubelet_retries=0, kubelet_retry_interval=0.0)
+    threads = [
+        threading.Thread(target=pm.get_pending_pods) for _ in range(10)
+    ]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert slow.calls == 1  # one remote list served all 10 callers
+
+
+def test_forced_refresh_rejects_stale_inflight_snapshot():
+    """A forced caller must NOT be satisfied by a list that started before
+    it asked."""
+    kube = FakeKubeClient("node-a")
+    pm = PodManager(kube, "node-a", kubelet_client=kube.as_kubelet(),
+                    cache_ttl=10.0, kubelet_retries=0,
+                    kubelet_retry_interval=0.0, apiserver_retries=0)
+    pm.get_pending_pods()          # warm cache (empty)
+    kube.add_pod(make_pod("late", 4, gpu_idx=0))
+    pods = pm.get_pending_pods(force_refresh=True)
+    assert [p["metadata"]["name"] for p in pods] == ["late"]
+
+
+# --------------------------------------------------------------------------- #
+# stub kubelet version contract
+# --------------------------------------------------------------------------- #
+
+def test_kubelet_rejects_wrong_api_version(tmp_socket_dir):
+    import grpc
+
+    from gpushare_amd.deviceplugin.stubkubelet import StubKubelet
+
+    kubelet = StubKubelet(tmp_socket_dir)
+    kubelet.start()
+    try:
+        with grpc.insecure_channel(f"unix://{kubelet.socket_path}") as ch:
+            register = ch.unary_unary(
+                api.METHOD_REGISTER,
+                request_serializer=lambda m: m.SerializeToString(),
+                response_deserializer=api.Empty.FromString,
+            )
+            with pytest.raises(grpc.RpcError) as err:
+                register(
+                    api.RegisterRequest(
+                        version="v1alpha1",
+                        endpoint="x.sock",
+                        resource_name="aliyun.com/gpu-mem",
+                    ),
+                    timeout=5,
+                )
+            assert err.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+    finally:
+        kubelet.stop()
