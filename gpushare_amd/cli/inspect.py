@@ -17,7 +17,6 @@ Deviation from the reference, deliberate: memory-unit inference used
 from __future__ import annotations
 
 import argparse
-import json
 import sys
 
 from .. import consts

@@ -18,6 +18,7 @@ Unlike the reference (Unhealthy is terminal — server.go:180 FIXME), a
 
 from __future__ import annotations
 
+import dataclasses
 import logging
 import threading
 import time
@@ -117,9 +118,7 @@ class AmdSmiSource:
                     if p in index_of_kfd
                 )
             )
-            resolved.append(
-                PhysicalGPU(**{**g.__dict__, "xgmi_peers": peers})
-            )
+            resolved.append(dataclasses.replace(g, xgmi_peers=peers))
         return resolved
 
     def devices(self) -> list[PhysicalGPU]:

@@ -11,7 +11,6 @@ from __future__ import annotations
 
 import queue
 import re
-import threading
 from typing import Iterable, Optional
 
 from . import HealthEvent, PhysicalGPU

@@ -83,12 +83,13 @@ class GPUShareExtender:
         name = podutils.pod_name(pod)
         if name:
             key = (podutils.pod_namespace(pod), name)
-            self._pod_cache[key] = (pod, time.monotonic())
-            if len(self._pod_cache) > 10_000:
-                cutoff = time.monotonic() - self.pod_cache_ttl
-                self._pod_cache = {
-                    k: v for k, v in self._pod_cache.items() if v[1] > cutoff
-                }
+            with self._lock:
+                self._pod_cache[key] = (pod, time.monotonic())
+                if len(self._pod_cache) > 10_000:
+                    cutoff = time.monotonic() - self.pod_cache_ttl
+                    self._pod_cache = {
+                        k: v for k, v in self._pod_cache.items() if v[1] > cutoff
+                    }
         request = podutils.gpu_memory_of_pod(pod)
         if request <= 0:
             return node_names
@@ -133,7 +134,8 @@ class GPUShareExtender:
 
     def cached_pod(self, namespace: str, name: str):
         """Pod captured at filter time, if still fresh (one-shot)."""
-        entry = self._pod_cache.pop((namespace, name), None)
+        with self._lock:
+            entry = self._pod_cache.pop((namespace, name), None)
         if entry is None:
             return None
         pod, t = entry

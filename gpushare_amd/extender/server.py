@@ -17,7 +17,6 @@ import json
 import threading
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
 
-from ..cluster import podutils
 from .core import GPUShareExtender
 
 

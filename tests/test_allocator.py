@@ -1,7 +1,5 @@
 """Allocator: match path, env contract, device injection, failure paths."""
 
-import pytest
-
 from gpushare_amd import consts
 from gpushare_amd.allocator import Allocator
 from gpushare_amd.cluster.kubeclient import FakeKubeClient

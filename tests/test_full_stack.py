@@ -8,7 +8,6 @@ Flow mirrors deploy/demo/binpack-1.yaml: four 72 GiB pods land on one
 """
 
 import io
-import json
 import threading
 
 import pytest

@@ -50,9 +50,7 @@ def test_kfd_topology_real():
         assert t.render_path, f"GPU {gpu_id} has no render node"
         assert os.path.exists(t.render_path), t.render_path
         assert t.vram_bytes > 250 << 30
-        # gfx950 target
-        assert t.gfx_target_version // 100 * 100 in (90500, 95000, 90000, 95001) or \
-            t.gfx_target_version > 0
+        assert t.gfx_target_version == 90500  # gfx950
 
 
 def test_source_resolves_render_and_uuid(source):

@@ -2,7 +2,7 @@
 
 import pytest
 
-from gpushare_amd import consts, metrics
+from gpushare_amd import metrics
 from gpushare_amd.allocator import Allocator
 from gpushare_amd.cluster.events import EventRecorder, NullRecorder
 from gpushare_amd.cluster.kubeclient import FakeKubeClient

@@ -1,6 +1,5 @@
 """Scale axes and concurrency edge cases (SURVEY §7 hard parts)."""
 
-import json
 import threading
 import time
 
