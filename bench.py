@@ -166,9 +166,7 @@ def run_wave(
         if rate_limiter is not None:
             rate_limiter.wait()
         spec = make_pod_spec(name, mem)
-        api._client.post(
-            "/api/v1/namespaces/default/pods", content=json.dumps(spec)
-        )
+        api.create_pod(spec)
         t0 = time.perf_counter()
         # real scheduler flow: filter (carries the full pod) then bind
         ext.filter(spec, [NODE])
@@ -176,7 +174,7 @@ def run_wave(
         lat_extender.append(time.perf_counter() - t0)
         if err:
             failed += 1
-            api._client.delete(f"/api/v1/namespaces/default/pods/{name}")
+            api.delete_pod("default", name)
             continue
         t0 = time.perf_counter()
         resp = plugin_client.allocate([grains[:mem]])
@@ -194,7 +192,7 @@ def run_wave(
     # delete + release (release stub built locally — the informer's delete
     # event carries the pod object, no extra apiserver GET needed)
     for name, mem in pods:
-        api._client.delete(f"/api/v1/namespaces/default/pods/{name}")
+        api.delete_pod("default", name)
         if name in placed:
             mem_placed, idx = placed[name]
             stub = make_pod_spec(name, mem_placed)
@@ -245,6 +243,11 @@ def main():
         )
         ext_url = ext_proc.stdout.readline().split()[1]
 
+        from gpushare_amd.cluster.informer import PodInformer
+
+        informer = PodInformer(RestKubeClient(base_url=api_url), NODE)
+        informer.start()
+        informer.wait_synced(timeout=10)
         pm = PodManager(
             RestKubeClient(base_url=api_url),
             NODE,
@@ -252,6 +255,7 @@ def main():
             query_kubelet=False,
             cache_ttl=0.05,
             apiserver_retries=0,
+            informer=informer,
         )
         sockdir = tempfile.mkdtemp(prefix="gpushare-bench-")
         plugin = GPUSharePlugin(

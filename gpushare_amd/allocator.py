@@ -155,35 +155,72 @@ class Allocator:
 
     # ------------------------------------------------------------------ #
     def _match_and_claim(self, req_units: int) -> Optional[dict]:
-        """Oldest unclaimed assumed pod with exact total-memory equality;
-        forced cache revalidation on miss (the TTL cache may trail the
-        extender), retried briefly: a concurrent Allocate may claim a pod
-        from our snapshot while its own (newer) pod is not yet in it — the
-        next snapshot covers that bind.  The listing itself runs outside
-        the claim lock."""
-        backoffs = (0.0, 0.0, 0.01, 0.04)
-        for attempt, force in enumerate((False, True, True, True)):
-            if backoffs[attempt]:
-                time.sleep(backoffs[attempt])
+        """Oldest unclaimed assumed pod with exact total-memory equality.
+
+        Informer path: re-check the in-memory store the instant a watch
+        event lands (condition wait, no fixed sleeps) within a ~50 ms
+        budget, then one authoritative remote list — a match is never
+        missed because the watch is behind.  List path (no informer):
+        forced cache revalidation with short backoffs, as the TTL cache
+        may trail the extender.  The listing runs outside the claim lock;
+        a concurrent Allocate may claim a pod from our snapshot while its
+        own (newer) pod is not yet in it — the next snapshot covers that
+        bind."""
+        informer = getattr(self.pods, "informer", None)
+        if informer is not None and informer.synced:
+            version = informer.version
+            pod = self._try_claim(self.pods.get_candidate_pods(), req_units)
+            if pod is not None:
+                return pod
+            deadline = time.monotonic() + 0.05
+            while True:
+                remaining = deadline - time.monotonic()
+                if remaining <= 0:
+                    break
+                version = informer.wait_newer(version, timeout=remaining)
+                pod = self._try_claim(
+                    self.pods.get_candidate_pods(), req_units
+                )
+                if pod is not None:
+                    return pod
+            attempts = ((0.0, True, True),)  # final: authoritative remote list
+        else:
+            attempts = (
+                (0.0, False, False),
+                (0.0, True, False),
+                (0.01, True, False),
+                (0.04, True, True),
+            )
+        for backoff, force, authoritative in attempts:
+            if backoff:
+                time.sleep(backoff)
             try:
-                candidates = self.pods.get_candidate_pods(force_refresh=force)
+                candidates = self.pods.get_candidate_pods(
+                    force_refresh=force, authoritative=authoritative
+                )
             except Exception as e:  # listing failed entirely
                 log.warning("failed to list candidate pods: %s", e)
                 return None
-            with self._lock:
-                now = time.monotonic()
-                for pod in candidates:
-                    if podutils.gpu_memory_of_pod(pod) != req_units:
-                        continue
-                    uid = podutils.pod_uid(pod)
-                    if self._claims.get(uid, 0.0) > now:
-                        continue  # being handled by a concurrent Allocate
-                    self._claims[uid] = now + self.claim_ttl
-                    if len(self._claims) > 10_000:
-                        self._claims = {
-                            u: t for u, t in self._claims.items() if t > now
-                        }
-                    return pod
+            pod = self._try_claim(candidates, req_units)
+            if pod is not None:
+                return pod
+        return None
+
+    def _try_claim(self, candidates: list, req_units: int) -> Optional[dict]:
+        with self._lock:
+            now = time.monotonic()
+            for pod in candidates:
+                if podutils.gpu_memory_of_pod(pod) != req_units:
+                    continue
+                uid = podutils.pod_uid(pod)
+                if self._claims.get(uid, 0.0) > now:
+                    continue  # being handled by a concurrent Allocate
+                self._claims[uid] = now + self.claim_ttl
+                if len(self._claims) > 10_000:
+                    self._claims = {
+                        u: t for u, t in self._claims.items() if t > now
+                    }
+                return pod
         return None
 
     def _unclaim(self, uid: str) -> None:

@@ -58,6 +58,9 @@ def parse_args(argv=None):
                    help="e.g. 1x8GiB: serve fake devices (no GPU needed)")
     p.add_argument("--no-inject", action="store_true",
                    help="skip /dev/kfd+/dev/dri DeviceSpec injection")
+    p.add_argument("--no-informer", action="store_true",
+                   help="disable the pod watch informer; every Allocate "
+                        "lists pods remotely (reference behavior)")
     p.add_argument("--metrics-port", type=int, default=0,
                    help="serve Prometheus /metrics on this port (0 = off)")
     p.add_argument("-v", "--verbose", action="count", default=0)
@@ -108,6 +111,7 @@ def main(argv=None) -> int:
             socket_dir=args.socket_dir,
             cache_ttl=args.cache_ttl,
             inject_devices=not args.no_inject,
+            use_informer=not args.no_informer,
         ),
     )
     mgr.install_signal_handlers()

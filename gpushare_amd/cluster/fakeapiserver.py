@@ -23,10 +23,11 @@ from __future__ import annotations
 
 import json
 import threading
-from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+from http.server import BaseHTTPRequestHandler
 from typing import Optional
 from urllib.parse import parse_qs, urlparse
 
+from .httpconn import TrackedThreadingHTTPServer
 from .kubeclient import FakeKubeClient, KubeError
 
 
@@ -57,6 +58,47 @@ class _Handler(BaseHTTPRequestHandler):
     def _body(self) -> dict:
         length = int(self.headers.get("Content-Length", 0))
         return json.loads(self.rfile.read(length)) if length else {}
+
+    def _stream_watch(self, field_selector: str) -> None:
+        """k8s `?watch=true`: chunked stream of newline-delimited watch
+        events, until the client hangs up.  Only a `spec.nodeName=` selector
+        is honored (what the informer asks for); phase transitions are the
+        consumer's business, as with a real informer."""
+        import queue
+
+        sel = dict(
+            kv.split("=", 1) for kv in field_selector.split(",") if "=" in kv
+        )
+        want_node = sel.get("spec.nodeName")
+        sub = self.store.watch_subscribe()
+        try:
+            self.send_response(200)
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Transfer-Encoding", "chunked")
+            self.end_headers()
+            while True:
+                try:
+                    item = sub.get(timeout=5.0)
+                    if item is None:       # server shutting down
+                        break
+                    node, line = item
+                except queue.Empty:
+                    # heartbeat chunk: detects a dead client, keeps NATs open
+                    self.wfile.write(b"1\r\n\n\r\n")
+                    self.wfile.flush()
+                    continue
+                if want_node and node != want_node:
+                    continue
+                payload = line + b"\n"
+                self.wfile.write(
+                    f"{len(payload):x}\r\n".encode() + payload + b"\r\n"
+                )
+                self.wfile.flush()
+        except (BrokenPipeError, ConnectionResetError, OSError):
+            pass
+        finally:
+            self.store.watch_unsubscribe(sub)
+            self.close_connection = True
 
     def _route(self, method: str) -> None:
         try:
@@ -93,6 +135,9 @@ class _Handler(BaseHTTPRequestHandler):
                 self._send(200, s.patch_node_status(name, self._body()))
                 return
         if rest == ["pods"] and method == "GET":
+            if q.get("watch", ["false"])[0] == "true":
+                self._stream_watch(q.get("fieldSelector", [""])[0])
+                return
             self._send_raw(200, s.list_pods_raw(q.get("fieldSelector", [""])[0]))
             return
         if rest[:1] == ["namespaces"] and len(rest) == 3 and rest[2] == "events":
@@ -142,7 +187,7 @@ class FakeApiServer:
     def __init__(self, store: Optional[FakeKubeClient] = None, port: int = 0):
         self.store = store or FakeKubeClient()
         handler = type("BoundHandler", (_Handler,), {"store": self.store})
-        self._httpd = ThreadingHTTPServer(("127.0.0.1", port), handler)
+        self._httpd = TrackedThreadingHTTPServer(("127.0.0.1", port), handler)
         self.port = self._httpd.server_port
         self._thread = threading.Thread(
             target=self._httpd.serve_forever, name="fake-apiserver", daemon=True
@@ -157,8 +202,10 @@ class FakeApiServer:
         return self
 
     def stop(self) -> None:
+        self.store.watch_close_all()
         self._httpd.shutdown()
         self._httpd.server_close()
+        self._httpd.stop_all_connections()
 
 
 def main(argv=None) -> int:

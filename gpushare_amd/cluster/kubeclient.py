@@ -23,6 +23,7 @@ import os
 import ssl
 import threading
 from typing import Optional
+from urllib.parse import urlencode
 
 import yaml
 
@@ -43,7 +44,7 @@ class ConflictError(KubeError):
 
 
 # --------------------------------------------------------------------------- #
-# Real clients (httpx)
+# Real clients (persistent keep-alive sessions, cluster/httpconn.py)
 # --------------------------------------------------------------------------- #
 
 class RestKubeClient:
@@ -57,18 +58,15 @@ class RestKubeClient:
         verify=None,
         timeout: float = 10.0,
     ):
-        import httpx
+        from .httpconn import HttpSession
 
         if base_url is None:
             base_url, token, verify = self._auto_config()
         headers = {"Accept": "application/json"}
         if token:
             headers["Authorization"] = f"Bearer {token}"
-        self._client = httpx.Client(
-            base_url=base_url,
-            headers=headers,
-            verify=verify if verify is not None else False,
-            timeout=timeout,
+        self._client = HttpSession(
+            base_url, headers=headers, verify=verify, timeout=timeout
         )
 
     @staticmethod
@@ -116,25 +114,34 @@ class RestKubeClient:
         return f"https://{host}:{port}", token, verify
 
     # -- verbs ---------------------------------------------------------------
-    def _check(self, resp):
-        if resp.status_code == 409:
-            raise ConflictError(resp.text)
-        if resp.status_code >= 400:
-            raise KubeError(resp.status_code, resp.text)
-        return resp.json()
+    _PATCH_HDRS = {"Content-Type": "application/strategic-merge-patch+json"}
+    _JSON_HDRS = {"Content-Type": "application/json"}
+
+    @staticmethod
+    def _raise(status: int, body: bytes):
+        text = body.decode(errors="replace")
+        if status == 409:
+            raise ConflictError(text)
+        raise KubeError(status, text)
+
+    def _check(self, status: int, body: bytes) -> dict:
+        if status >= 400:
+            self._raise(status, body)
+        return json.loads(body)
 
     def get_node(self, name: str) -> dict:
-        return self._check(self._client.get(f"/api/v1/nodes/{name}"))
+        return self._check(*self._client.request("GET", f"/api/v1/nodes/{name}"))
 
     def list_nodes(self) -> dict:
-        return self._check(self._client.get("/api/v1/nodes"))
+        return self._check(*self._client.request("GET", "/api/v1/nodes"))
 
     def patch_node_status(self, name: str, patch: dict) -> dict:
         return self._check(
-            self._client.patch(
+            *self._client.request(
+                "PATCH",
                 f"/api/v1/nodes/{name}/status",
-                content=json.dumps(patch),
-                headers={"Content-Type": "application/strategic-merge-patch+json"},
+                body=json.dumps(patch).encode(),
+                headers=self._PATCH_HDRS,
             )
         )
 
@@ -142,35 +149,69 @@ class RestKubeClient:
         path = (
             f"/api/v1/namespaces/{namespace}/pods" if namespace else "/api/v1/pods"
         )
-        params = {"fieldSelector": field_selector} if field_selector else {}
-        return self._check(self._client.get(path, params=params))
+        if field_selector:
+            path += "?" + urlencode({"fieldSelector": field_selector})
+        return self._check(*self._client.request("GET", path))
 
     def get_pod(self, namespace: str, name: str) -> dict:
         return self._check(
-            self._client.get(f"/api/v1/namespaces/{namespace}/pods/{name}")
+            *self._client.request(
+                "GET", f"/api/v1/namespaces/{namespace}/pods/{name}"
+            )
         )
+
+    def create_pod(self, pod: dict, namespace: str = "default") -> None:
+        status, body = self._client.request(
+            "POST",
+            f"/api/v1/namespaces/{namespace}/pods",
+            body=json.dumps(pod).encode(),
+            headers=self._JSON_HDRS,
+        )
+        if status >= 400:
+            self._raise(status, body)
+
+    def delete_pod(self, namespace: str, name: str) -> None:
+        status, body = self._client.request(
+            "DELETE", f"/api/v1/namespaces/{namespace}/pods/{name}"
+        )
+        if status >= 400:
+            self._raise(status, body)
 
     def patch_pod(self, namespace: str, name: str, patch: dict,
                   parse: bool = True):
-        resp = self._client.patch(
+        status, body = self._client.request(
+            "PATCH",
             f"/api/v1/namespaces/{namespace}/pods/{name}",
-            content=json.dumps(patch),
-            headers={"Content-Type": "application/strategic-merge-patch+json"},
+            body=json.dumps(patch).encode(),
+            headers=self._PATCH_HDRS,
         )
-        if parse:
-            return self._check(resp)
-        if resp.status_code == 409:
-            raise ConflictError(resp.text)
-        if resp.status_code >= 400:
-            raise KubeError(resp.status_code, resp.text)
-        return None
+        if status >= 400:
+            self._raise(status, body)
+        return json.loads(body) if parse else None
+
+    def watch_pods_stream(self, field_selector: str = ""):
+        """Open a k8s watch on pods (chunked stream); returns (conn, resp).
+        The caller reads newline-delimited watch events from ``resp`` and
+        closes ``conn`` when done (used by cluster.informer.PodInformer)."""
+        params = {"watch": "true"}
+        if field_selector:
+            params["fieldSelector"] = field_selector
+        conn, resp = self._client.stream(
+            "GET", "/api/v1/pods?" + urlencode(params), timeout=30.0
+        )
+        if resp.status >= 400:
+            body = resp.read()
+            conn.close()
+            self._raise(resp.status, body)
+        return conn, resp
 
     def create_event(self, namespace: str, event: dict) -> dict:
         return self._check(
-            self._client.post(
+            *self._client.request(
+                "POST",
                 f"/api/v1/namespaces/{namespace}/events",
-                content=json.dumps(event),
-                headers={"Content-Type": "application/json"},
+                body=json.dumps(event).encode(),
+                headers=self._JSON_HDRS,
             )
         )
 
@@ -189,8 +230,9 @@ class KubeletClient:
         port: int = 10250,
         token: Optional[str] = None,
         timeout: float = 10.0,
+        scheme: str = "https",
     ):
-        import httpx
+        from .httpconn import HttpSession
 
         headers = {"Accept": "application/json"}
         if token is None:
@@ -199,18 +241,18 @@ class KubeletClient:
                 token = open(token_path).read().strip()
         if token:
             headers["Authorization"] = f"Bearer {token}"
-        self._client = httpx.Client(
-            base_url=f"https://{address}:{port}",
+        self._client = HttpSession(
+            f"{scheme}://{address}:{port}",
             headers=headers,
             verify=False,
             timeout=timeout,
         )
 
     def get_node_running_pods(self) -> dict:
-        resp = self._client.get("/pods/")
-        if resp.status_code >= 400:
-            raise KubeError(resp.status_code, resp.text)
-        return resp.json()
+        status, body = self._client.request("GET", "/pods/")
+        if status >= 400:
+            raise KubeError(status, body.decode(errors="replace"))
+        return json.loads(body)
 
     def close(self) -> None:
         self._client.close()
@@ -240,6 +282,44 @@ class FakeKubeClient:
         self.fail_next_pod_patches = 0   # inject N consecutive 409s
         self.patch_count = 0
         self.list_count = 0
+        self.watch_count = 0
+        self._watchers: list = []        # SimpleQueue per open watch stream
+
+    # -- watch feed (k8s `?watch=true` semantics for the fake) ---------------
+    def watch_subscribe(self):
+        import queue
+
+        q = queue.SimpleQueue()
+        with self._lock:
+            self._watchers.append(q)
+            self.watch_count += 1
+        return q
+
+    def watch_unsubscribe(self, q) -> None:
+        with self._lock:
+            if q in self._watchers:
+                self._watchers.remove(q)
+
+    def watch_close_all(self) -> None:
+        """Terminate every open watch stream (server shutdown): a ``None``
+        sentinel makes the handler close its connection."""
+        with self._lock:
+            for q in self._watchers:
+                q.put(None)
+            self._watchers.clear()
+
+    def _publish(self, event_type: str, key: tuple) -> None:
+        """Queue a watch event; caller holds self._lock.  The event carries
+        the pod's encoded bytes (DELETED: the final state before removal)."""
+        if not self._watchers:
+            return
+        node = self.pods[key].get("spec", {}).get("nodeName", "")
+        line = (
+            b'{"type":"' + event_type.encode() + b'","object":'
+            + self._pod_bytes[key] + b"}"
+        )
+        for q in self._watchers:
+            q.put((node, line))
 
     def _bump(self, obj: dict) -> None:
         self._rv += 1
@@ -255,14 +335,19 @@ class FakeKubeClient:
             pod["metadata"].setdefault("namespace", "default")
             pod["metadata"].setdefault("uid", f"uid-{key[0]}-{key[1]}")
             self._bump(pod)
+            added = key not in self.pods
             self.pods[key] = pod
             self._reencode(key)
+            self._publish("ADDED" if added else "MODIFIED", key)
             return pod
 
     def delete_pod(self, namespace: str, name: str) -> None:
         with self._lock:
-            self.pods.pop((namespace, name), None)
-            self._pod_bytes.pop((namespace, name), None)
+            key = (namespace, name)
+            if key in self._pod_bytes:
+                self._publish("DELETED", key)
+            self.pods.pop(key, None)
+            self._pod_bytes.pop(key, None)
 
     # -- apiserver verbs -----------------------------------------------------
     def get_node(self, name: str) -> dict:
@@ -332,6 +417,7 @@ class FakeKubeClient:
                 pod["metadata"].setdefault("annotations", {}).update(anns)
             self._bump(pod)
             self._reencode(key)
+            self._publish("MODIFIED", key)
             return json.loads(json.dumps(pod)) if parse else None
 
     # -- raw-bytes accessors (fakeapiserver hot path: the per-pod JSON is

@@ -49,9 +49,11 @@ class PodManager:
         kubelet_retry_interval: float = KUBELET_RETRY_INTERVAL,
         apiserver_retries: int = APISERVER_RETRIES,
         apiserver_retry_interval: float = APISERVER_RETRY_INTERVAL,
+        informer=None,
     ):
         self.kube = kube_client
         self.kubelet = kubelet_client
+        self.informer = informer      # cluster.informer.PodInformer (optional)
         self.node_name = node_name
         self.query_kubelet = query_kubelet and kubelet_client is not None
         self.cache_ttl = cache_ttl
@@ -100,15 +102,29 @@ class PodManager:
             f"failed to get Pods assigned to node {self.node_name}: {last_err}"
         )
 
-    def get_pending_pods(self, force_refresh: bool = False) -> list:
+    def get_pending_pods(
+        self, force_refresh: bool = False, authoritative: bool = False
+    ) -> list:
         """Pending pods on this node, deduped by UID (reference:
-        getPendingPodsInNode, podmanager.go:162-212), TTL-cached.
+        getPendingPodsInNode, podmanager.go:162-212).
+
+        Fast path: the informer's in-memory store — zero remote calls
+        (watch events land push-style; the store is already node-scoped
+        and uid-keyed).  ``authoritative=True`` bypasses the informer for
+        a direct remote list — the allocator's final matching attempt uses
+        it so scheduling never depends on watch freshness.
 
         Remote lists are single-flighted: N concurrent callers (N parallel
         Allocates missing the cache at once) trigger ONE remote list; a
         forced caller is satisfied only by a list that *started after it
         asked* (a list already in flight may predate the caller's own
         bind)."""
+        if (
+            not authoritative
+            and self.informer is not None
+            and self.informer.synced
+        ):
+            return self.informer.pending_pods()
         entered = time.monotonic()
         with self._cache_lock:
             while True:
@@ -178,10 +194,14 @@ class PodManager:
                         consts.ENV_ASSIGNED_FLAG
                     ] = "true"
 
-    def get_candidate_pods(self, force_refresh: bool = False) -> list:
+    def get_candidate_pods(
+        self, force_refresh: bool = False, authoritative: bool = False
+    ) -> list:
         """Assumed-but-unassigned pods, oldest assume-time first (FIFO
         fairness; reference: getCandidatePods, podmanager.go:215-262)."""
-        pods = self.get_pending_pods(force_refresh=force_refresh)
+        pods = self.get_pending_pods(
+            force_refresh=force_refresh, authoritative=authoritative
+        )
         candidates = [p for p in pods if podutils.is_assumed_pod(p)]
         candidates.sort(key=podutils.assume_time_from_annotation)
         return candidates

@@ -15,8 +15,9 @@ from __future__ import annotations
 
 import json
 import threading
-from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+from http.server import BaseHTTPRequestHandler
 
+from ..cluster.httpconn import TrackedThreadingHTTPServer
 from .core import GPUShareExtender
 
 
@@ -88,7 +89,7 @@ class _Handler(BaseHTTPRequestHandler):
 class ExtenderServer:
     def __init__(self, extender: GPUShareExtender, port: int = 0):
         handler = type("BoundHandler", (_Handler,), {"extender": extender})
-        self._httpd = ThreadingHTTPServer(("127.0.0.1", port), handler)
+        self._httpd = TrackedThreadingHTTPServer(("127.0.0.1", port), handler)
         self.port = self._httpd.server_port
         self._thread = threading.Thread(
             target=self._httpd.serve_forever, name="extender", daemon=True
@@ -105,41 +106,48 @@ class ExtenderServer:
     def stop(self) -> None:
         self._httpd.shutdown()
         self._httpd.server_close()
+        self._httpd.stop_all_connections()
 
 
 class ExtenderClient:
     """Client used by churn generators (and by a scheduler integration)."""
 
     def __init__(self, url: str, timeout: float = 10.0):
-        import httpx
+        from ..cluster.httpconn import HttpSession
 
-        self._client = httpx.Client(base_url=url, timeout=timeout)
+        self._client = HttpSession(url, timeout=timeout)
+
+    def _post(self, path: str, obj: dict) -> dict:
+        status, body = self._client.request(
+            "POST", path, body=json.dumps(obj).encode(),
+            headers={"Content-Type": "application/json"},
+        )
+        if status >= 400:
+            raise RuntimeError(
+                f"extender {path}: HTTP {status}: {body.decode(errors='replace')}"
+            )
+        return json.loads(body)
 
     def filter(self, pod: dict, node_names: list[str]) -> list[str]:
-        r = self._client.post(
+        return self._post(
             "/gpushare-scheduler/filter",
-            json={"Pod": pod, "NodeNames": node_names},
-        )
-        r.raise_for_status()
-        return r.json()["NodeNames"]
+            {"Pod": pod, "NodeNames": node_names},
+        )["NodeNames"]
 
     def bind(self, namespace: str, name: str, node: str) -> str:
-        r = self._client.post(
+        return self._post(
             "/gpushare-scheduler/bind",
-            json={"PodNamespace": namespace, "PodName": name, "Node": node},
-        )
-        r.raise_for_status()
-        return r.json().get("Error", "")
+            {"PodNamespace": namespace, "PodName": name, "Node": node},
+        ).get("Error", "")
 
     def release(self, pod: dict, node: str = "") -> None:
-        self._client.post(
-            "/gpushare-scheduler/release", json={"Pod": pod, "Node": node}
-        ).raise_for_status()
+        self._post("/gpushare-scheduler/release", {"Pod": pod, "Node": node})
 
     def packing(self) -> dict:
-        r = self._client.get("/gpushare-scheduler/packing")
-        r.raise_for_status()
-        return r.json()
+        status, body = self._client.request("GET", "/gpushare-scheduler/packing")
+        if status >= 400:
+            raise RuntimeError(f"extender packing: HTTP {status}")
+        return json.loads(body)
 
     def close(self) -> None:
         self._client.close()

@@ -46,6 +46,8 @@ class ManagerOptions:
     cache_ttl: float = 0.2
     inject_devices: bool = True
     coredump_dir: str = "/etc/kubernetes"
+    use_informer: bool = True    # watch-based pending-pod cache (informer.py);
+                                 # falls back to LIST paths while unsynced
 
 
 class SharedGPUManager:
@@ -64,6 +66,7 @@ class SharedGPUManager:
         self.opt = options or ManagerOptions()
         self.plugin: GPUSharePlugin = None
         self.health: HealthMonitor = None
+        self.pod_informer = None
         self._signals: "queue.Queue[int]" = queue.Queue()
         self._stop = threading.Event()
 
@@ -72,12 +75,17 @@ class SharedGPUManager:
         gpus = self.source.devices()
         if not gpus:
             raise RuntimeError("no GPUs found; refusing to serve")
+        if self.opt.use_informer and self.pod_informer is None:
+            from .cluster.informer import PodInformer
+
+            self.pod_informer = PodInformer(self.kube, self.node_name).start()
         pm = PodManager(
             self.kube,
             self.node_name,
             kubelet_client=self.kubelet,
             query_kubelet=self.opt.query_kubelet,
             cache_ttl=self.opt.cache_ttl,
+            informer=self.pod_informer,
         )
         pm.patch_gpu_count(len(gpus))
         allocator = Allocator(
@@ -161,6 +169,9 @@ class SharedGPUManager:
                     self._stop.set()
         finally:
             self._stop_plugin()
+            if self.pod_informer is not None:
+                self.pod_informer.stop()
+                self.pod_informer = None
             watcher.close()
 
     def shutdown(self) -> None:
