@@ -75,7 +75,12 @@ class SharedGPUManager:
         gpus = self.source.devices()
         if not gpus:
             raise RuntimeError("no GPUs found; refusing to serve")
-        if self.opt.use_informer and self.pod_informer is None:
+        if (
+            self.opt.use_informer
+            and self.pod_informer is None
+            # in-memory fakes (tests, mock mode) have no watch stream
+            and hasattr(self.kube, "watch_pods_stream")
+        ):
             from .cluster.informer import PodInformer
 
             self.pod_informer = PodInformer(self.kube, self.node_name).start()
