@@ -235,10 +235,12 @@ def main():
             "allocatable": {consts.RESOURCE_COUNT: str(len(gpus)),
                             consts.RESOURCE_NAME: str(sum(units))},
         }})
+        # --no-watch: the bench releases explicitly (synchronous wave
+        # teardown); the daemon's informer auto-release would race it
         ext_proc = subprocess.Popen(
             [sys.executable, "-m", "gpushare_amd.extender",
              "--api-url", api_url, "--port", "0",
-             "--resync-interval", "3600"],
+             "--resync-interval", "3600", "--no-watch"],
             stdout=subprocess.PIPE, text=True, cwd=repo,
         )
         ext_url = ext_proc.stdout.readline().split()[1]

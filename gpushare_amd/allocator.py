@@ -125,11 +125,11 @@ class Allocator:
             list_t = time.perf_counter() - tl
             if pod is not None:
                 uid = podutils.pod_uid(pod)
-                gpu = self._gpu_for_pod(pod)
-                if gpu is None:
+                gpus = self._gpus_for_pod(pod)
+                if not gpus:
                     self._unclaim(uid)
                     return self._err_response(request, req_units)
-                resp = self._build_response(request, req_units, gpu)
+                resp = self._build_response(request, req_units, gpus)
                 tp = time.perf_counter()
                 patched = self.pods.mark_assigned(pod)
                 patch_t = time.perf_counter() - tp
@@ -142,7 +142,7 @@ class Allocator:
                 # single-GPU fast path (allocate.go:151-178)
                 gpu = next(iter(self.gpus.values()))
                 ok = True
-                return self._build_response(request, req_units, gpu)
+                return self._build_response(request, req_units, [gpu])
             log.warning(
                 "invalid allocation request: %d %s cannot be matched to "
                 "an assumed pod",
@@ -227,7 +227,24 @@ class Allocator:
         with self._lock:
             self._claims.pop(uid, None)
 
-    def _gpu_for_pod(self, pod: dict) -> Optional[PhysicalGPU]:
+    def _gpus_for_pod(self, pod: dict) -> list[PhysicalGPU]:
+        """Bound GPU(s): the multi-GPU allocation-map annotation when the
+        extender split the pod over several xGMI-adjacent GPUs, else the
+        single-index annotation (reference protocol)."""
+        split = podutils.gpu_split_from_pod(pod)
+        if split:
+            gpus = []
+            for idx in sorted(split):
+                gpu = self.gpus.get(idx)
+                if gpu is None:
+                    log.warning(
+                        "allocation map points at GPU %d which does not "
+                        "exist",
+                        idx,
+                    )
+                    return []
+                gpus.append(gpu)
+            return gpus
         idx = podutils.gpu_id_from_annotation(pod)
         if idx < 0:
             log.warning(
@@ -236,33 +253,44 @@ class Allocator:
                 podutils.pod_name(pod),
                 consts.ENV_RESOURCE_INDEX,
             )
-            return None
+            return []
         gpu = self.gpus.get(idx)
         if gpu is None:
             log.warning("annotation points at GPU %d which does not exist", idx)
-        return gpu
+            return []
+        return [gpu]
 
     # ------------------------------------------------------------------ #
     def _build_response(
-        self, request, req_units: int, gpu: PhysicalGPU
+        self, request, req_units: int, gpus: list[PhysicalGPU]
     ) -> "api.AllocateResponse":
-        dev_units = gpu.mem_units(self.unit)
-        rocr_id = gpu.extras.get("rocr_uuid") or str(gpu.index)
+        """Container envs + AMD device nodes for the bound GPU(s).  For a
+        multi-GPU placement every container sees the whole set (ROCr
+        ordinal space is consistent across co-located containers, so RCCL
+        picks the direct xGMI paths); ``ALIYUN_COM_GPU_MEM_IDX`` stays the
+        single primary index for reference-protocol readers."""
+        primary = gpus[0]
+        dev_units = primary.mem_units(self.unit)
+        rocr_id = ",".join(
+            g.extras.get("rocr_uuid") or str(g.index) for g in gpus
+        )
+        hip_visible = ",".join(str(i) for i in range(len(gpus)))
         responses = api.AllocateResponse()
         for cr in request.container_requests:
             c = responses.container_responses.add()
             c.envs[consts.ENV_ROCR_VISIBLE] = rocr_id
-            c.envs[consts.ENV_HIP_VISIBLE] = "0"
-            c.envs[consts.ENV_RESOURCE_INDEX] = str(gpu.index)
+            c.envs[consts.ENV_HIP_VISIBLE] = hip_visible
+            c.envs[consts.ENV_RESOURCE_INDEX] = str(primary.index)
             c.envs[consts.ENV_RESOURCE_BY_POD] = str(req_units)
             c.envs[consts.ENV_RESOURCE_BY_CONTAINER] = str(len(cr.devicesIDs))
             c.envs[consts.ENV_RESOURCE_BY_DEV] = str(dev_units)
             if self.disable_isolation:
                 c.envs[consts.ENV_CGPU_DISABLE] = "true"
             if self.inject_devices:
-                for host_path in filter(
-                    None, (consts.DEV_KFD, gpu.render_path, gpu.card_path)
-                ):
+                paths = [consts.DEV_KFD]
+                for g in gpus:
+                    paths.extend((g.render_path, g.card_path))
+                for host_path in filter(None, paths):
                     spec = c.devices.add()
                     spec.container_path = host_path
                     spec.host_path = host_path

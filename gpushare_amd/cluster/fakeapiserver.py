@@ -131,6 +131,9 @@ class _Handler(BaseHTTPRequestHandler):
             if method == "GET" and len(rest) == 2:
                 self._send(200, s.get_node(name))
                 return
+            if method == "PATCH" and len(rest) == 2:
+                self._send(200, s.patch_node(name, self._body()))
+                return
             if method == "PATCH" and rest[2:] == ["status"]:
                 self._send(200, s.patch_node_status(name, self._body()))
                 return

@@ -47,12 +47,14 @@ class PodInformer:
     def __init__(
         self,
         kube,                       # RestKubeClient
-        node_name: str,
+        node_name: Optional[str] = None,   # None ⇒ watch all nodes' pods
         resync_interval: float = 300.0,
         reconnect_backoff: float = 1.0,
+        on_event=None,              # callback(etype, pod) after each apply
     ):
         self.kube = kube
         self.node_name = node_name
+        self.on_event = on_event
         self.resync_interval = resync_interval
         self.reconnect_backoff = reconnect_backoff
         self._store: dict[str, dict] = {}
@@ -125,18 +127,21 @@ class PodInformer:
     # sync loop
     # ------------------------------------------------------------------ #
     def _run(self) -> None:
+        selector = (
+            f"spec.nodeName={self.node_name}" if self.node_name else ""
+        )
         backoff = self.reconnect_backoff
         while not self._stop.is_set():
             conn = None
             try:
                 conn, resp = self.kube.watch_pods_stream(
-                    field_selector=f"spec.nodeName={self.node_name}"
+                    field_selector=selector
                 )
                 # LIST after the watch is open: an event raced between the
                 # two waits in the stream's socket buffer and is reconciled
                 # by resourceVersion when read below — nothing is lost.
                 snapshot = self.kube.list_pods(
-                    field_selector=f"spec.nodeName={self.node_name}"
+                    field_selector=selector
                 ).get("items", [])
                 with self._lock:
                     self._store = {_uid(p): p for p in snapshot}
@@ -154,14 +159,20 @@ class PodInformer:
                     line = line.strip()
                     if line:
                         evt = json.loads(line)
-                        self._apply(evt.get("type", ""), evt.get("object", {}))
+                        etype, obj = evt.get("type", ""), evt.get("object", {})
+                        self._apply(etype, obj)
                         self.events_seen += 1
+                        if self.on_event is not None:
+                            try:
+                                self.on_event(etype, obj)
+                            except Exception as e:  # noqa: BLE001
+                                log.warning("informer on_event failed: %s", e)
                     if (
                         self.resync_interval > 0
                         and time.monotonic() - last_list > self.resync_interval
                     ):
                         snapshot = self.kube.list_pods(
-                            field_selector=f"spec.nodeName={self.node_name}"
+                            field_selector=selector
                         ).get("items", [])
                         self._merge_snapshot(snapshot)
                         self.relists += 1

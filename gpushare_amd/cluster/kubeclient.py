@@ -145,6 +145,17 @@ class RestKubeClient:
             )
         )
 
+    def patch_node(self, name: str, patch: dict) -> dict:
+        """Strategic-merge patch of node metadata (topology annotation)."""
+        return self._check(
+            *self._client.request(
+                "PATCH",
+                f"/api/v1/nodes/{name}",
+                body=json.dumps(patch).encode(),
+                headers=self._PATCH_HDRS,
+            )
+        )
+
     def list_pods(self, field_selector: str = "", namespace: str = "") -> dict:
         path = (
             f"/api/v1/namespaces/{namespace}/pods" if namespace else "/api/v1/pods"
@@ -370,6 +381,21 @@ class FakeKubeClient:
                 node["status"].setdefault(sect, {}).update(
                     patch.get("status", {}).get(sect, {})
                 )
+            self._bump(node)
+            return json.loads(json.dumps(node))
+
+    def patch_node(self, name: str, patch: dict) -> dict:
+        with self._lock:
+            if name not in self.nodes:
+                raise KubeError(404, f"node {name} not found")
+            node = self.nodes[name]
+            md = patch.get("metadata", {})
+            if md.get("annotations"):
+                node["metadata"].setdefault("annotations", {}).update(
+                    md["annotations"]
+                )
+            if md.get("labels"):
+                node["metadata"].setdefault("labels", {}).update(md["labels"])
             self._bump(node)
             return json.loads(json.dumps(node))
 

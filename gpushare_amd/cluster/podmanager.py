@@ -247,6 +247,28 @@ class PodManager:
         }
         self.kube.patch_node_status(self.node_name, patch)
 
+    def patch_topology(self, gpus: list, unit: str = consts.GIB) -> None:
+        """Publish real per-GPU capacities + xGMI adjacency as a node
+        annotation (MI355X-native: lets the extender place multi-GPU pods
+        on link-adjacent sets and drop the reference's homogeneous-node
+        assumption, nvidia.go:70-72)."""
+        import json as _json
+
+        topo = {
+            "unit": unit,
+            "per_gpu_units": [g.mem_units(unit) for g in gpus],
+            "xgmi": [sorted(g.xgmi_peers) for g in gpus],
+        }
+        patch = {
+            "metadata": {
+                "annotations": {consts.ANN_NODE_TOPOLOGY: _json.dumps(topo)}
+            }
+        }
+        try:
+            self.kube.patch_node(self.node_name, patch)
+        except KubeError as e:
+            log.warning("failed to publish topology annotation: %s", e)
+
     def isolation_disabled(self) -> bool:
         """Node label `cgpu.disable.isolation=true` check (reference:
         disableCGPUIsolationOrNot, podmanager.go:59-72)."""

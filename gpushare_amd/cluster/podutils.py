@@ -96,6 +96,23 @@ def allocation_map_from_annotation(pod: dict) -> Optional[dict]:
         return None
 
 
+def gpu_split_from_pod(pod: dict) -> Optional[dict[int, int]]:
+    """Merged per-GPU placement {gpu_idx: units} from the allocation-map
+    annotation (summed across containers); None if the annotation is
+    absent/invalid.  A multi-GPU pod's authoritative placement record."""
+    alloc = allocation_map_from_annotation(pod)
+    if not alloc:
+        return None
+    merged: dict[int, int] = {}
+    try:
+        for per_gpu in alloc.values():
+            for idx_s, units in per_gpu.items():
+                merged[int(idx_s)] = merged.get(int(idx_s), 0) + int(units)
+    except (TypeError, ValueError, AttributeError):
+        return None
+    return merged or None
+
+
 # --------------------------------------------------------------------------- #
 # predicates
 # --------------------------------------------------------------------------- #

@@ -46,7 +46,19 @@ ENV_RESOURCE_ASSIGN_TIME = "ALIYUN_COM_GPU_MEM_ASSIGN_TIME"
 
 # Per-container allocation map written by newer scheduler-extender versions
 # (reference: cmd/inspect/nodeinfo.go:244-271): JSON {container: {gpuIdx: mem}}.
+# Also the record of a multi-GPU placement (a pod whose gpu-mem request spans
+# several xGMI-linked MI355X) — the inspect CLI and the plugin's Allocate
+# both read it.
 ANN_GPUSHARE_ALLOCATION = "scheduler.framework.gpushare.allocation"
+
+# Node annotation published by the plugin: real per-GPU capacities and the
+# xGMI adjacency from KFD topology, JSON
+# {"unit": "GiB", "per_gpu_units": [..], "xgmi": [[peer idx..], ..]}.
+# The scheduler extender uses it to (a) place multi-GPU pods on
+# link-adjacent GPU sets so RCCL inside the containers runs over xGMI, and
+# (b) drop the homogeneous-capacity assumption the reference bakes in
+# (nvidia.go:70-72 reads GPU0's memory for every device).
+ANN_NODE_TOPOLOGY = "gpushare.amd.com/topology"
 
 # ---------------------------------------------------------------------------
 # AMD-native container injection (replaces NVIDIA_VISIBLE_DEVICES; on ROCm the

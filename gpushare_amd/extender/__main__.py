@@ -21,15 +21,34 @@ log = logging.getLogger("extender")
 
 
 def discover_nodes(kube, extender) -> int:
+    """Register shared-GPU nodes.  Prefer the plugin-published topology
+    annotation (real per-GPU capacities + xGMI adjacency for multi-GPU
+    placement); fall back to the uniform total/count split the reference
+    assumes (nvidia.go:70-72)."""
+    import json
+
     n = 0
     for node in kube.list_nodes().get("items", []):
         alloc = node.get("status", {}).get("allocatable", {})
         count = int(alloc.get(consts.RESOURCE_COUNT, 0) or 0)
         total = int(alloc.get(consts.RESOURCE_NAME, 0) or 0)
-        if count > 0 and total > 0:
-            name = node["metadata"]["name"]
-            extender.register_node(name, [total // count] * count)
-            n += 1
+        if not (count > 0 and total > 0):
+            continue
+        name = node["metadata"]["name"]
+        per_gpu, xgmi = [total // count] * count, None
+        raw = (node["metadata"].get("annotations") or {}).get(
+            consts.ANN_NODE_TOPOLOGY
+        )
+        if raw:
+            try:
+                topo = json.loads(raw)
+                if len(topo.get("per_gpu_units", [])) == count:
+                    per_gpu = [int(u) for u in topo["per_gpu_units"]]
+                    xgmi = topo.get("xgmi")
+            except (ValueError, TypeError, KeyError) as e:
+                log.warning("bad topology annotation on %s: %s", name, e)
+        extender.register_node(name, per_gpu, xgmi=xgmi)
+        n += 1
     return n
 
 
@@ -37,6 +56,9 @@ def main(argv=None) -> int:
     p = argparse.ArgumentParser(prog="gpushare-scheduler-extender")
     p.add_argument("--port", type=int, default=32766)
     p.add_argument("--resync-interval", type=float, default=30.0)
+    p.add_argument("--no-watch", action="store_true",
+                   help="disable the pod watch (auto-release of deleted "
+                        "pods' reservations); rely on resync only")
     p.add_argument("--api-url", default=None, help=argparse.SUPPRESS)
     args = p.parse_args(argv)
     logging.basicConfig(level=logging.INFO, stream=sys.stderr)
@@ -45,6 +67,13 @@ def main(argv=None) -> int:
     extender = GPUShareExtender(kube, resync_interval=args.resync_interval)
     n = discover_nodes(kube, extender)
     extender.resync()
+
+    informer = None
+    if not args.no_watch:
+        from .core import make_auto_release_informer
+
+        informer = make_auto_release_informer(kube, extender).start()
+
     log.info("serving binpack extender for %d shared-GPU node(s)", n)
     server = ExtenderServer(extender, port=args.port)
     server._thread.daemon = False
@@ -56,6 +85,8 @@ def main(argv=None) -> int:
             discover_nodes(kube, extender)
             extender.resync()
     except KeyboardInterrupt:
+        if informer is not None:
+            informer.stop()
         server.stop()
     return 0
 

@@ -9,6 +9,7 @@ reference: podutils.go:78-119).
 
 from __future__ import annotations
 
+import json
 import logging
 import threading
 import time
@@ -34,12 +35,18 @@ class GPUShareExtender:
         # filters before binding) — saves the bind-path GET to the apiserver
         self._pod_cache: dict[tuple, tuple] = {}
         self.pod_cache_ttl = 30.0
+        # idempotence: the webhook `release` and the informer's DELETED
+        # event may both fire for one pod — releasing twice would free
+        # units another pod holds
+        self._released: dict[str, float] = {}
+        self.released_ttl = 60.0
 
     # ------------------------------------------------------------------ #
     # state sync
     # ------------------------------------------------------------------ #
-    def register_node(self, node: str, per_gpu_units: list[int]) -> None:
-        self.state.set_node(node, per_gpu_units)
+    def register_node(self, node: str, per_gpu_units: list[int],
+                      xgmi: Optional[list] = None) -> None:
+        self.state.set_node(node, per_gpu_units, xgmi=xgmi)
 
     def resync(self, nodes: Optional[list[str]] = None) -> None:
         """Rebuild the ledger from pod annotations (source of truth)."""
@@ -61,10 +68,16 @@ class GPUShareExtender:
                 anns = podutils.annotations(pod)
                 if consts.ENV_RESOURCE_ASSUME_TIME not in anns:
                     continue
+                split = podutils.gpu_split_from_pod(pod)
+                if split:
+                    for idx, units in split.items():
+                        if 0 <= idx < len(allocated):
+                            allocated[idx] += units
+                    continue
                 idx = podutils.gpu_id_from_annotation(pod)
                 if 0 <= idx < len(allocated):
                     allocated[idx] += mem
-            self.state.set_node(node, st.per_gpu_units, allocated)
+            self.state.set_node(node, st.per_gpu_units, allocated, st.xgmi)
         self._last_resync = time.monotonic()
 
     def _maybe_resync(self) -> None:
@@ -96,37 +109,72 @@ class GPUShareExtender:
         return self.state.filter_nodes(request, node_names)
 
     def assume(self, pod: dict, node: str) -> Optional[int]:
-        """Webhook `bind` body: best-fit a GPU, write the annotation triple.
-        Returns the GPU index or None if the node cannot fit the pod."""
+        """Webhook `bind` body: best-fit GPU(s), write the annotation triple.
+        A request no single GPU can hold is split over the smallest
+        most-xGMI-connected set (binpack.best_fit_multi) and recorded in the
+        per-container allocation-map annotation the plugin and inspect CLI
+        read.  Returns the primary (lowest) GPU index, or None if the node
+        cannot fit the pod."""
         request = podutils.gpu_memory_of_pod(pod)
         if request <= 0:
             return None
-        idx = self.state.assume(node, request)
-        if idx is None:
+        split = self.state.assume_multi(node, request)
+        if split is None:
             self.rejected += 1
             return None
+        idx = min(split)
         ns = podutils.pod_namespace(pod)
         name = podutils.pod_name(pod)
-        patch = {
-            "metadata": {
-                "annotations": {
-                    consts.ENV_RESOURCE_INDEX: str(idx),
-                    consts.ENV_RESOURCE_ASSUME_TIME: str(time.time_ns()),
-                    consts.ENV_ASSIGNED_FLAG: "false",
-                }
-            }
+        anns = {
+            consts.ENV_RESOURCE_INDEX: str(idx),
+            consts.ENV_RESOURCE_ASSUME_TIME: str(time.time_ns()),
+            consts.ENV_ASSIGNED_FLAG: "false",
         }
+        if len(split) > 1:
+            container = next(
+                (
+                    c.get("name", "main")
+                    for c in pod.get("spec", {}).get("containers", [])
+                    if int(
+                        c.get("resources", {})
+                        .get("limits", {})
+                        .get(consts.RESOURCE_NAME, 0)
+                    )
+                    > 0
+                ),
+                "main",
+            )
+            anns[consts.ANN_GPUSHARE_ALLOCATION] = json.dumps(
+                {container: {str(i): u for i, u in sorted(split.items())}}
+            )
+        patch = {"metadata": {"annotations": anns}}
         try:
             self.kube.patch_pod(ns, name, patch, parse=False)
         except Exception as e:  # noqa: BLE001
             log.warning("assume patch failed for %s/%s: %s", ns, name, e)
-            self.state.release(node, idx, request)
+            self.state.release_multi(node, split)
             return None
         self.assumed += 1
         return idx
 
     def release(self, pod: dict, node: str) -> None:
-        """Informer delete-event path: return the pod's reservation."""
+        """Return the pod's reservation (webhook `release` or the
+        informer's DELETED event — idempotent across both)."""
+        uid = podutils.pod_uid(pod)
+        if uid:  # uid-less caller stubs are always honored
+            now = time.monotonic()
+            with self._lock:
+                if self._released.get(uid, 0.0) > now:
+                    return
+                self._released[uid] = now + self.released_ttl
+                if len(self._released) > 10_000:
+                    self._released = {
+                        u: t for u, t in self._released.items() if t > now
+                    }
+        split = podutils.gpu_split_from_pod(pod)
+        if split:
+            self.state.release_multi(node, split)
+            return
         request = podutils.gpu_memory_of_pod(pod)
         idx = podutils.gpu_id_from_annotation(pod)
         if request > 0 and idx >= 0:
@@ -145,3 +193,20 @@ class GPUShareExtender:
 
     def packing(self) -> dict:
         return self.state.packing()
+
+
+def make_auto_release_informer(kube, extender: GPUShareExtender):
+    """All-nodes pod watch that returns a deleted gpushare pod's
+    reservation immediately (production flow: kubectl delete → watch
+    DELETED → release; idempotent with the webhook release path).  The
+    caller starts/stops the returned informer."""
+    from ..cluster.informer import PodInformer
+
+    def _on_event(etype: str, pod: dict) -> None:
+        if etype != "DELETED" or podutils.gpu_memory_of_pod(pod) <= 0:
+            return
+        node = pod.get("spec", {}).get("nodeName", "")
+        if node:
+            extender.release(pod, node)
+
+    return PodInformer(kube, None, on_event=_on_event)

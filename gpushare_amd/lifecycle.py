@@ -93,6 +93,7 @@ class SharedGPUManager:
             informer=self.pod_informer,
         )
         pm.patch_gpu_count(len(gpus))
+        pm.patch_topology(gpus, unit=self.opt.memory_unit)
         allocator = Allocator(
             gpus,
             pm,
