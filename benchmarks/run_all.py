@@ -94,6 +94,14 @@ def main() -> int:
                    "--pods-per-gpu", "4", "--rate", "10"],
             ranks=8,
         ),
+        # MI355X-native extension (not a reference config): pods larger
+        # than one GPU, split over xGMI-adjacent pairs by the extender
+        "6_multigpu_span_5x400": dict(
+            extra=["--gpus", "8", "--mock", "8x288GiB", "--steps", steps,
+                   "--warmup", "2", "--pods-per-gpu", "5",
+                   "--pod-gib", "400"],
+            ranks=1,
+        ),
     }
     for name, cfg in configs.items():
         t0 = time.time()

@@ -40,6 +40,10 @@ class GPUShareExtender:
         # units another pod holds
         self._released: dict[str, float] = {}
         self.released_ttl = 60.0
+        # placement record from assume time: release() falls back to it
+        # when the caller's pod copy lacks the allocation-map annotation
+        # (e.g. a delete-event stub); pruned on resync against live pods
+        self._placements: dict[tuple, tuple] = {}   # (ns,name) -> (node, split)
 
     # ------------------------------------------------------------------ #
     # state sync
@@ -78,6 +82,13 @@ class GPUShareExtender:
                 if 0 <= idx < len(allocated):
                     allocated[idx] += mem
             self.state.set_node(node, st.per_gpu_units, allocated, st.xgmi)
+        live = {
+            (podutils.pod_namespace(p), podutils.pod_name(p)) for p in pods
+        }
+        with self._lock:
+            self._placements = {
+                k: v for k, v in self._placements.items() if k in live
+            }
         self._last_resync = time.monotonic()
 
     def _maybe_resync(self) -> None:
@@ -154,6 +165,8 @@ class GPUShareExtender:
             log.warning("assume patch failed for %s/%s: %s", ns, name, e)
             self.state.release_multi(node, split)
             return None
+        with self._lock:
+            self._placements[(ns, name)] = (node, split)
         self.assumed += 1
         return idx
 
@@ -171,7 +184,12 @@ class GPUShareExtender:
                     self._released = {
                         u: t for u, t in self._released.items() if t > now
                     }
+        key = (podutils.pod_namespace(pod), podutils.pod_name(pod))
+        with self._lock:
+            recorded = self._placements.pop(key, None)
         split = podutils.gpu_split_from_pod(pod)
+        if split is None and recorded is not None:
+            node, split = recorded[0] or node, recorded[1]
         if split:
             self.state.release_multi(node, split)
             return

@@ -285,3 +285,37 @@ class TestAutoRelease:
             inf.stop()
         finally:
             server.stop()
+
+
+class TestPlacementRecordFallback:
+    def test_release_stub_without_map_frees_full_split(self):
+        """A release carrying only the primary-index annotation (e.g. a
+        scheduler's delete stub) must still free the full multi-GPU split,
+        via the extender's placement record from assume time."""
+        kube = FakeKubeClient(node_name=NODE)
+        ext = GPUShareExtender(kube, resync_interval=3600)
+        ext.register_node(NODE, [16, 16], xgmi=[[1], [0]])
+        pod = make_pod("big", node=NODE, mem=24)
+        del pod["metadata"]["annotations"]
+        kube.add_pod(pod)
+        assert ext.assume(pod, NODE) == 0
+        assert ext.state.packing()["allocated_units"] == 24
+
+        stub = make_pod("big", node=NODE, mem=24, gpu_idx=0)
+        stub["metadata"].pop("uid", None)  # caller stub, no uid / no map
+        ext.release(stub, NODE)
+        assert ext.state.packing()["allocated_units"] == 0
+        assert ext.state.packing()["per_node"][NODE] == [0, 0]
+
+    def test_resync_prunes_stale_placements(self):
+        kube = FakeKubeClient(node_name=NODE)
+        ext = GPUShareExtender(kube, resync_interval=3600)
+        ext.register_node(NODE, [16, 16], xgmi=[[1], [0]])
+        pod = make_pod("gone", node=NODE, mem=24)
+        del pod["metadata"]["annotations"]
+        kube.add_pod(pod)
+        ext.assume(pod, NODE)
+        kube.delete_pod("default", "gone")
+        ext.resync()
+        assert ext._placements == {}
+        assert ext.state.packing()["allocated_units"] == 0
