@@ -26,6 +26,8 @@ import threading
 import time
 from typing import Optional
 
+from .. import metrics
+
 log = logging.getLogger(__name__)
 
 
@@ -149,6 +151,7 @@ class PodInformer:
                     self._cv.notify_all()
                 self.relists += 1
                 self._synced.set()
+                metrics.observe_informer_state(True)
                 backoff = self.reconnect_backoff
                 last_list = time.monotonic()
 
@@ -162,6 +165,7 @@ class PodInformer:
                         etype, obj = evt.get("type", ""), evt.get("object", {})
                         self._apply(etype, obj)
                         self.events_seen += 1
+                        metrics.observe_informer_event()
                         if self.on_event is not None:
                             try:
                                 self.on_event(etype, obj)
@@ -182,6 +186,7 @@ class PodInformer:
                     break
                 self._synced.clear()
                 self.reconnects += 1
+                metrics.observe_informer_state(False, reconnected=True)
                 log.warning(
                     "pod informer stream failed (%s); reconnecting in %.1fs",
                     e,

@@ -56,6 +56,15 @@ if AVAILABLE:
     LISTANDWATCH_SENDS = Counter(
         "gpushare_listandwatch_sends_total", "ListAndWatch payloads sent"
     )
+    INFORMER_EVENTS = Counter(
+        "gpushare_informer_events_total", "Pod watch events applied"
+    )
+    INFORMER_RECONNECTS = Counter(
+        "gpushare_informer_reconnects_total", "Pod watch stream reconnects"
+    )
+    INFORMER_SYNCED = Gauge(
+        "gpushare_informer_synced", "Pod informer synced (1) / degraded (0)"
+    )
 
 
 def observe_allocate(total_s: float, list_s: float, patch_s: float, ok: bool) -> None:
@@ -82,6 +91,19 @@ def observe_inventory(n_devices: int) -> None:
 def observe_law_send() -> None:
     if AVAILABLE:
         LISTANDWATCH_SENDS.inc()
+
+
+def observe_informer_event() -> None:
+    if AVAILABLE:
+        INFORMER_EVENTS.inc()
+
+
+def observe_informer_state(synced: bool, reconnected: bool = False) -> None:
+    if not AVAILABLE:
+        return
+    INFORMER_SYNCED.set(1 if synced else 0)
+    if reconnected:
+        INFORMER_RECONNECTS.inc()
 
 
 def serve(port: int) -> Optional[object]:
