@@ -158,3 +158,49 @@ def test_health_flip_visible_to_kubelet_full_stack(stack):
     source.inject_health_event(HealthEvent(gpu_index=1, healthy=True))
     devices = client.wait_for_update(min_updates=3, timeout=15)
     assert all(h == consts.HEALTHY for h in devices.values())
+
+
+def test_multigpu_span_full_stack(stack):
+    """A 400 GiB pod (> one 288 GiB GPU) through the whole stack: the
+    extender must split it over both GPUs (allocation-map annotation), the
+    plugin's real gRPC Allocate must inject BOTH render nodes, and the
+    inspect CLI must show the per-GPU split."""
+    api, kube, kubelet, ext, extender, _source = stack
+    # register with xGMI adjacency as the daemon's topology annotation would
+    extender.register_node(NODE, [288, 288], xgmi=[[1], [0]])
+    client = kubelet.wait_for_plugin(consts.RESOURCE_NAME, timeout=15)
+    devices = client.wait_for_devices(min_count=576, timeout=15)
+    grains = sorted(devices)
+
+    api.store.add_pod(_gpu_pod("span", 400))
+    assert ext.filter(api.store.get_pod("default", "span"), [NODE]) == [NODE]
+    assert ext.bind("default", "span", NODE) == ""
+
+    anns = api.store.get_pod("default", "span")["metadata"]["annotations"]
+    alloc = anns[consts.ANN_GPUSHARE_ALLOCATION]
+    import json as _json
+
+    merged = {
+        int(i): u
+        for per in _json.loads(alloc).values()
+        for i, u in per.items()
+    }
+    assert sum(merged.values()) == 400 and set(merged) == {0, 1}
+
+    resp = client.allocate([grains[:400]])
+    c = resp.container_responses[0]
+    assert c.envs[consts.ENV_HIP_VISIBLE] == "0,1"
+    assert len(c.envs[consts.ENV_ROCR_VISIBLE].split(",")) == 2
+    paths = {d.host_path for d in c.devices}
+    assert {"/dev/dri/renderD128", "/dev/dri/renderD129"} <= paths
+
+    # inspect shows the split across both GPUs
+    api.store.pods[("default", "span")]["status"]["phase"] = "Running"
+    api.store._reencode(("default", "span"))
+    kube.patch_node_status(
+        NODE, {"status": {"allocatable": {consts.RESOURCE_NAME: "576"}}}
+    )
+    out = io.StringIO()
+    assert insp.main(["--api-url", api.url], kube=None, out=out) == 0
+    text = out.getvalue()
+    assert "400/576" in text
