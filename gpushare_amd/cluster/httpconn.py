@@ -139,6 +139,12 @@ class HttpSession:
                 resp = conn.getresponse()
                 data = resp.read()  # must drain before conn reuse
                 return resp.status, data
+            except socket.timeout:
+                # the request reached a live-but-slow server: retrying
+                # could double-apply a non-idempotent verb — surface it
+                conn.close()
+                self._local.conn = None
+                raise
             except (
                 http.client.RemoteDisconnected,
                 http.client.BadStatusLine,
@@ -146,7 +152,6 @@ class HttpSession:
                 BrokenPipeError,
                 ConnectionResetError,
                 ConnectionRefusedError,
-                socket.timeout,
                 ssl.SSLEOFError,
             ):
                 conn.close()

@@ -203,12 +203,16 @@ class RestKubeClient:
     def watch_pods_stream(self, field_selector: str = ""):
         """Open a k8s watch on pods (chunked stream); returns (conn, resp).
         The caller reads newline-delimited watch events from ``resp`` and
-        closes ``conn`` when done (used by cluster.informer.PodInformer)."""
-        params = {"watch": "true"}
+        closes ``conn`` when done (used by cluster.informer.PodInformer).
+
+        ``timeoutSeconds=300`` asks the apiserver to close the watch
+        cleanly at 5 min (the informer reconnects); the socket timeout sits
+        just above so a dead server is still detected on a quiet node."""
+        params = {"watch": "true", "timeoutSeconds": "300"}
         if field_selector:
             params["fieldSelector"] = field_selector
         conn, resp = self._client.stream(
-            "GET", "/api/v1/pods?" + urlencode(params), timeout=30.0
+            "GET", "/api/v1/pods?" + urlencode(params), timeout=330.0
         )
         if resp.status >= 400:
             body = resp.read()
