@@ -217,3 +217,33 @@ class TestPodManagerInformerPath:
             assert anns[consts.ENV_ASSIGNED_FLAG] == "true"
         finally:
             inf.stop()
+
+
+class TestDegradedMode:
+    def test_allocate_works_with_unsynced_informer(self, api):
+        """Watch endpoint unreachable (informer never syncs) ⇒ the
+        allocator transparently uses the remote-list path — the informer
+        is an accelerator, not a correctness dependency."""
+
+        class NeverSyncs:
+            synced = False
+            version = 0
+
+            def pending_pods(self):  # pragma: no cover
+                raise AssertionError("must not be consulted while unsynced")
+
+        gpus = MockSource.from_spec("2x16GiB").devices()
+        pm = PodManager(
+            RestKubeClient(base_url=api.url),
+            NODE,
+            kubelet_client=None,
+            query_kubelet=False,
+            informer=NeverSyncs(),
+            cache_ttl=0.0,
+            apiserver_retries=0,
+        )
+        alloc = Allocator(gpus, pm)
+        api.store.add_pod(make_pod("p1", node=NODE, mem=8, gpu_idx=1))
+        resp = alloc.allocate(_request([8]))
+        envs = resp.container_responses[0].envs
+        assert envs[consts.ENV_RESOURCE_INDEX] == "1"
