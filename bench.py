@@ -407,6 +407,14 @@ def main():
         print(json.dumps(result), flush=True)
 
     # --- teardown ---------------------------------------------------------
+    # the measurement is already printed; a wedged teardown (grpc stream,
+    # informer reconnect loop, subprocess) must never hang the run
+    import threading as _threading
+
+    _watchdog = _threading.Timer(20.0, lambda: os._exit(0))
+    _watchdog.daemon = True
+    _watchdog.start()
+    sys.stdout.flush()
     plugin_client.close()
     ext.close()
     api.close()
@@ -414,11 +422,13 @@ def main():
         dist.barrier()
         dist.destroy_process_group()
     if rank == 0:
+        informer.stop()
         plugin.stop()
         ext_proc.terminate()
         api_proc.terminate()
         ext_proc.wait(timeout=5)
         api_proc.wait(timeout=5)
+    os._exit(0)
 
 
 if __name__ == "__main__":
