@@ -8,207 +8,157 @@ Wraps :class:`FakeKubeClient`'s in-memory store with the REST surface that
   scheduler-extender, and N churn-generator ranks all talk to over
   127.0.0.1 — the same process topology as a real node.
 
+Served by :class:`~gpushare_amd.cluster.fasthttp.FastHTTPServer` (minimal
+keep-alive HTTP) so the harness apiserver is not the measured ceiling —
+a real kube-apiserver is far faster than a Python stdlib HTTP stack.
+
 Endpoints (subset of the k8s API the framework uses):
   GET    /api/v1/nodes/{name}
+  PATCH  /api/v1/nodes/{name}            (metadata: topology annotation)
   PATCH  /api/v1/nodes/{name}/status
-  GET    /api/v1/pods?fieldSelector=...
+  GET    /api/v1/pods?fieldSelector=...[&watch=true]
   GET    /api/v1/namespaces/{ns}/pods/{name}
   PATCH  /api/v1/namespaces/{ns}/pods/{name}
   POST   /api/v1/namespaces/{ns}/pods          (bench: pod creation)
   DELETE /api/v1/namespaces/{ns}/pods/{name}   (bench: pod deletion)
+  POST   /api/v1/namespaces/{ns}/events
   GET    /pods/                                 (kubelet read-only view)
 """
 
 from __future__ import annotations
 
 import json
-import threading
-from http.server import BaseHTTPRequestHandler
+import queue
 from typing import Optional
 from urllib.parse import parse_qs, urlparse
 
-from .httpconn import TrackedThreadingHTTPServer
+from .fasthttp import FastHTTPServer
 from .kubeclient import FakeKubeClient, KubeError
 
 
-class _Handler(BaseHTTPRequestHandler):
-    protocol_version = "HTTP/1.1"
-    disable_nagle_algorithm = True
-    store: FakeKubeClient = None  # set by server factory
-
-    def log_message(self, fmt, *args):  # quiet
-        pass
-
-    # -- helpers -------------------------------------------------------------
-    def _send_raw(self, code: int, body: bytes) -> None:
-        self.send_response(code)
-        self.send_header("Content-Type", "application/json")
-        self.send_header("Content-Length", str(len(body)))
-        self.end_headers()
-        self.wfile.write(body)
-
-    def _send(self, code: int, obj) -> None:
-        body = json.dumps(obj).encode()
-        self.send_response(code)
-        self.send_header("Content-Type", "application/json")
-        self.send_header("Content-Length", str(len(body)))
-        self.end_headers()
-        self.wfile.write(body)
-
-    def _body(self) -> dict:
-        length = int(self.headers.get("Content-Length", 0))
-        return json.loads(self.rfile.read(length)) if length else {}
-
-    def _stream_watch(self, field_selector: str) -> None:
-        """k8s `?watch=true`: chunked stream of newline-delimited watch
-        events, until the client hangs up.  Only a `spec.nodeName=` selector
-        is honored (what the informer asks for); phase transitions are the
-        consumer's business, as with a real informer."""
-        import queue
-
-        sel = dict(
-            kv.split("=", 1) for kv in field_selector.split(",") if "=" in kv
-        )
-        want_node = sel.get("spec.nodeName")
-        sub = self.store.watch_subscribe()
-        try:
-            self.send_response(200)
-            self.send_header("Content-Type", "application/json")
-            self.send_header("Transfer-Encoding", "chunked")
-            self.end_headers()
-            while True:
-                try:
-                    item = sub.get(timeout=5.0)
-                    if item is None:       # server shutting down
-                        break
-                    node, line = item
-                except queue.Empty:
-                    # heartbeat chunk: detects a dead client, keeps NATs open
-                    self.wfile.write(b"1\r\n\n\r\n")
-                    self.wfile.flush()
-                    continue
-                if want_node and node != want_node:
-                    continue
-                payload = line + b"\n"
-                self.wfile.write(
-                    f"{len(payload):x}\r\n".encode() + payload + b"\r\n"
-                )
-                self.wfile.flush()
-        except (BrokenPipeError, ConnectionResetError, OSError):
-            pass
-        finally:
-            self.store.watch_unsubscribe(sub)
-            self.close_connection = True
-
-    def _route(self, method: str) -> None:
-        try:
-            self._route_inner(method)
-        except KubeError as e:
-            self._send(e.status, {"kind": "Status", "message": str(e)})
-        except Exception as e:  # noqa: BLE001
-            self._send(500, {"kind": "Status", "message": str(e)})
-
-    def _route_inner(self, method: str) -> None:
-        u = urlparse(self.path)
-        parts = [p for p in u.path.split("/") if p]
-        q = parse_qs(u.query)
-        s = self.store
-
-        if method == "GET" and parts == ["pods"]:
-            # kubelet read-only view
-            self._send_raw(200, s.kubelet_pods_raw())
-            return
-        if parts[:2] != ["api", "v1"]:
-            self._send(404, {"message": "not found"})
-            return
-        rest = parts[2:]
-
-        if rest == ["nodes"] and method == "GET":
-            self._send(200, s.list_nodes())
-            return
-        if rest[:1] == ["nodes"] and len(rest) >= 2:
-            name = rest[1]
-            if method == "GET" and len(rest) == 2:
-                self._send(200, s.get_node(name))
-                return
-            if method == "PATCH" and len(rest) == 2:
-                self._send(200, s.patch_node(name, self._body()))
-                return
-            if method == "PATCH" and rest[2:] == ["status"]:
-                self._send(200, s.patch_node_status(name, self._body()))
-                return
-        if rest == ["pods"] and method == "GET":
-            if q.get("watch", ["false"])[0] == "true":
-                self._stream_watch(q.get("fieldSelector", [""])[0])
-                return
-            self._send_raw(200, s.list_pods_raw(q.get("fieldSelector", [""])[0]))
-            return
-        if rest[:1] == ["namespaces"] and len(rest) == 3 and rest[2] == "events":
-            if method == "POST":
-                self._send(201, s.create_event(rest[1], self._body()))
-                return
-        if rest[:1] == ["namespaces"] and len(rest) >= 3 and rest[2] == "pods":
-            ns = rest[1]
-            if len(rest) == 3 and method == "POST":
-                pod = self._body()
-                pod.setdefault("metadata", {})["namespace"] = ns
-                self._send(201, s.add_pod(pod))
-                return
-            if len(rest) == 3 and method == "GET":
-                self._send_raw(200, s.list_pods_raw(namespace=ns))
-                return
-            if len(rest) == 4:
-                name = rest[3]
-                if method == "GET":
-                    self._send_raw(200, s.get_pod_raw(ns, name))
-                    return
-                if method == "PATCH":
-                    self._send(200, s.patch_pod(ns, name, self._body()))
-                    return
-                if method == "DELETE":
-                    s.delete_pod(ns, name)
-                    self._send(200, {"kind": "Status", "status": "Success"})
-                    return
-        self._send(404, {"message": f"no route {method} {u.path}"})
-
-    def do_GET(self):
-        self._route("GET")
-
-    def do_POST(self):
-        self._route("POST")
-
-    def do_PATCH(self):
-        self._route("PATCH")
-
-    def do_DELETE(self):
-        self._route("DELETE")
+def _json_bytes(obj) -> bytes:
+    return json.dumps(obj).encode()
 
 
 class FakeApiServer:
-    """Threaded HTTP apiserver around a FakeKubeClient store."""
+    """HTTP apiserver around a FakeKubeClient store (fasthttp-served)."""
 
     def __init__(self, store: Optional[FakeKubeClient] = None, port: int = 0):
         self.store = store or FakeKubeClient()
-        handler = type("BoundHandler", (_Handler,), {"store": self.store})
-        self._httpd = TrackedThreadingHTTPServer(("127.0.0.1", port), handler)
-        self.port = self._httpd.server_port
-        self._thread = threading.Thread(
-            target=self._httpd.serve_forever, name="fake-apiserver", daemon=True
-        )
+        self._httpd = FastHTTPServer(self._handle, port=port)
+        self.port = self._httpd.port
 
     @property
     def url(self) -> str:
         return f"http://127.0.0.1:{self.port}"
 
     def start(self) -> "FakeApiServer":
-        self._thread.start()
+        self._httpd.start()
         return self
 
     def stop(self) -> None:
         self.store.watch_close_all()
-        self._httpd.shutdown()
-        self._httpd.server_close()
-        self._httpd.stop_all_connections()
+        self._httpd.stop()
+
+    # ------------------------------------------------------------------ #
+    def _handle(self, method: str, path: str, raw_body: bytes):
+        try:
+            return self._route(method, path, raw_body)
+        except KubeError as e:
+            return e.status, _json_bytes(
+                {"kind": "Status", "message": str(e)}
+            )
+        except Exception as e:  # noqa: BLE001
+            return 500, _json_bytes({"kind": "Status", "message": str(e)})
+
+    def _route(self, method: str, path: str, raw_body: bytes):
+        u = urlparse(path)
+        parts = [p for p in u.path.split("/") if p]
+        q = parse_qs(u.query)
+        s = self.store
+        body = lambda: json.loads(raw_body) if raw_body else {}  # noqa: E731
+
+        if method == "GET" and parts == ["pods"]:
+            return 200, s.kubelet_pods_raw()          # kubelet read-only view
+        if parts[:2] != ["api", "v1"]:
+            return 404, b'{"message":"not found"}'
+        rest = parts[2:]
+
+        if rest == ["nodes"] and method == "GET":
+            return 200, _json_bytes(s.list_nodes())
+        if rest[:1] == ["nodes"] and len(rest) >= 2:
+            name = rest[1]
+            if method == "GET" and len(rest) == 2:
+                return 200, _json_bytes(s.get_node(name))
+            if method == "PATCH" and len(rest) == 2:
+                return 200, _json_bytes(s.patch_node(name, body()))
+            if method == "PATCH" and rest[2:] == ["status"]:
+                return 200, _json_bytes(s.patch_node_status(name, body()))
+        if rest == ["pods"] and method == "GET":
+            selector = q.get("fieldSelector", [""])[0]
+            if q.get("watch", ["false"])[0] == "true":
+                return "stream", self._watch_stream(selector)
+            return 200, s.list_pods_raw(selector)
+        if rest[:1] == ["namespaces"] and len(rest) == 3 and rest[2] == "events":
+            if method == "POST":
+                return 201, _json_bytes(s.create_event(rest[1], body()))
+        if rest[:1] == ["namespaces"] and len(rest) >= 3 and rest[2] == "pods":
+            ns = rest[1]
+            if len(rest) == 3 and method == "POST":
+                pod = body()
+                pod.setdefault("metadata", {})["namespace"] = ns
+                return 201, _json_bytes(s.add_pod(pod))
+            if len(rest) == 3 and method == "GET":
+                return 200, s.list_pods_raw(namespace=ns)
+            if len(rest) == 4:
+                name = rest[3]
+                if method == "GET":
+                    return 200, s.get_pod_raw(ns, name)
+                if method == "PATCH":
+                    return 200, _json_bytes(s.patch_pod(ns, name, body()))
+                if method == "DELETE":
+                    s.delete_pod(ns, name)
+                    return 200, b'{"kind":"Status","status":"Success"}'
+        return 404, _json_bytes({"message": f"no route {method} {u.path}"})
+
+    # ------------------------------------------------------------------ #
+    def _watch_stream(self, field_selector: str):
+        """k8s `?watch=true`: chunked stream of newline-delimited watch
+        events until the client hangs up or the store shuts down.  Only a
+        `spec.nodeName=` selector is honored (what the informer asks for);
+        phase transitions are the consumer's business, as with a real
+        informer."""
+        sel = dict(
+            kv.split("=", 1) for kv in field_selector.split(",") if "=" in kv
+        )
+        want_node = sel.get("spec.nodeName")
+        store = self.store
+
+        def run(conn) -> None:
+            sub = store.watch_subscribe()
+            try:
+                while True:
+                    try:
+                        item = sub.get(timeout=5.0)
+                        if item is None:        # server shutting down
+                            break
+                        node, line = item
+                    except queue.Empty:
+                        # heartbeat chunk: detects a dead client
+                        conn.sendall(b"1\r\n\n\r\n")
+                        continue
+                    if want_node and node != want_node:
+                        continue
+                    payload = line + b"\n"
+                    conn.sendall(
+                        f"{len(payload):x}\r\n".encode() + payload + b"\r\n"
+                    )
+            except (BrokenPipeError, ConnectionResetError, OSError):
+                pass
+            finally:
+                store.watch_unsubscribe(sub)
+
+        return run
 
 
 def main(argv=None) -> int:
@@ -217,7 +167,6 @@ def main(argv=None) -> int:
     Used by bench.py so the apiserver is its own process (as in a real
     cluster) instead of sharing the plugin's GIL."""
     import argparse
-    import sys
     import time
 
     p = argparse.ArgumentParser(prog="gpushare-fake-apiserver")

@@ -75,9 +75,7 @@ def main(argv=None) -> int:
         informer = make_auto_release_informer(kube, extender).start()
 
     log.info("serving binpack extender for %d shared-GPU node(s)", n)
-    server = ExtenderServer(extender, port=args.port)
-    server._thread.daemon = False
-    server.start()
+    server = ExtenderServer(extender, port=args.port).start()
     print(f"READY {server.url}", flush=True)
     try:
         while True:

@@ -14,99 +14,77 @@ at the same URL prefix the gpushare scheduler-extender uses, plus a
 from __future__ import annotations
 
 import json
-import threading
-from http.server import BaseHTTPRequestHandler
 
-from ..cluster.httpconn import TrackedThreadingHTTPServer
+from ..cluster.fasthttp import FastHTTPServer
 from .core import GPUShareExtender
 
 
-class _Handler(BaseHTTPRequestHandler):
-    protocol_version = "HTTP/1.1"
-    disable_nagle_algorithm = True
-    extender: GPUShareExtender = None  # bound by server factory
-
-    def log_message(self, fmt, *args):
-        pass
-
-    def _send(self, code: int, obj) -> None:
-        body = json.dumps(obj).encode()
-        self.send_response(code)
-        self.send_header("Content-Type", "application/json")
-        self.send_header("Content-Length", str(len(body)))
-        self.end_headers()
-        self.wfile.write(body)
-
-    def _body(self) -> dict:
-        length = int(self.headers.get("Content-Length", 0))
-        return json.loads(self.rfile.read(length)) if length else {}
-
-    def do_GET(self):
-        if self.path == "/gpushare-scheduler/packing":
-            self._send(200, self.extender.packing())
-        else:
-            self._send(404, {"message": "not found"})
-
-    def do_POST(self):
-        try:
-            body = self._body()
-            if self.path == "/gpushare-scheduler/filter":
-                pod = body.get("Pod") or {}
-                names = body.get("NodeNames") or [
-                    n.get("metadata", {}).get("name")
-                    for n in (body.get("Nodes") or {}).get("Items", [])
-                ]
-                ok = self.extender.filter(pod, [n for n in names if n])
-                self._send(
-                    200, {"NodeNames": ok, "FailedNodes": {}, "Error": ""}
-                )
-            elif self.path == "/gpushare-scheduler/bind":
-                ns = body.get("PodNamespace", "default")
-                name = body.get("PodName", "")
-                node = body.get("Node", "")
-                pod = self.extender.cached_pod(ns, name)
-                if pod is None:
-                    pod = self.extender.kube.get_pod(ns, name)
-                idx = self.extender.assume(pod, node)
-                if idx is None:
-                    self._send(
-                        200,
-                        {"Error": f"no GPU on {node} fits pod {ns}/{name}"},
-                    )
-                else:
-                    self._send(200, {"Error": ""})
-            elif self.path == "/gpushare-scheduler/release":
-                pod = body.get("Pod") or {}
-                node = body.get("Node") or pod.get("spec", {}).get("nodeName", "")
-                self.extender.release(pod, node)
-                self._send(200, {"Error": ""})
-            else:
-                self._send(404, {"message": "not found"})
-        except Exception as e:  # noqa: BLE001
-            self._send(500, {"Error": str(e)})
-
-
 class ExtenderServer:
+    """Webhook server on the minimal fasthttp stack (the filter/bind RTT is
+    inside the measured scheduling path)."""
+
     def __init__(self, extender: GPUShareExtender, port: int = 0):
-        handler = type("BoundHandler", (_Handler,), {"extender": extender})
-        self._httpd = TrackedThreadingHTTPServer(("127.0.0.1", port), handler)
-        self.port = self._httpd.server_port
-        self._thread = threading.Thread(
-            target=self._httpd.serve_forever, name="extender", daemon=True
-        )
+        self.extender = extender
+        self._httpd = FastHTTPServer(self._handle, port=port)
+        self.port = self._httpd.port
+        self._thread = None  # kept for API compat (daemon flag setters)
 
     @property
     def url(self) -> str:
         return f"http://127.0.0.1:{self.port}"
 
     def start(self) -> "ExtenderServer":
-        self._thread.start()
+        self._httpd.start()
         return self
 
     def stop(self) -> None:
-        self._httpd.shutdown()
-        self._httpd.server_close()
-        self._httpd.stop_all_connections()
+        self._httpd.stop()
+
+    # ------------------------------------------------------------------ #
+    def _handle(self, method: str, path: str, raw_body: bytes):
+        try:
+            return self._route(method, path, raw_body)
+        except Exception as e:  # noqa: BLE001
+            return 500, json.dumps({"Error": str(e)}).encode()
+
+    def _route(self, method: str, path: str, raw_body: bytes):
+        ext = self.extender
+        if method == "GET":
+            if path == "/gpushare-scheduler/packing":
+                return 200, json.dumps(ext.packing()).encode()
+            return 404, b'{"message":"not found"}'
+        if method != "POST":
+            return 404, b'{"message":"not found"}'
+        body = json.loads(raw_body) if raw_body else {}
+        if path == "/gpushare-scheduler/filter":
+            pod = body.get("Pod") or {}
+            names = body.get("NodeNames") or [
+                n.get("metadata", {}).get("name")
+                for n in (body.get("Nodes") or {}).get("Items", [])
+            ]
+            ok = ext.filter(pod, [n for n in names if n])
+            return 200, json.dumps(
+                {"NodeNames": ok, "FailedNodes": {}, "Error": ""}
+            ).encode()
+        if path == "/gpushare-scheduler/bind":
+            ns = body.get("PodNamespace", "default")
+            name = body.get("PodName", "")
+            node = body.get("Node", "")
+            pod = ext.cached_pod(ns, name)
+            if pod is None:
+                pod = ext.kube.get_pod(ns, name)
+            idx = ext.assume(pod, node)
+            if idx is None:
+                return 200, json.dumps(
+                    {"Error": f"no GPU on {node} fits pod {ns}/{name}"}
+                ).encode()
+            return 200, b'{"Error": ""}'
+        if path == "/gpushare-scheduler/release":
+            pod = body.get("Pod") or {}
+            node = body.get("Node") or pod.get("spec", {}).get("nodeName", "")
+            ext.release(pod, node)
+            return 200, b'{"Error": ""}'
+        return 404, b'{"message":"not found"}'
 
 
 class ExtenderClient:
