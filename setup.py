@@ -35,7 +35,7 @@ setup(
     packages=find_packages(include=["gpushare_amd*"]),
     package_data={"gpushare_amd": ["*.so"]},
     python_requires=">=3.9",
-    install_requires=["grpcio", "protobuf", "httpx", "pyyaml", "pybind11"],
+    install_requires=["grpcio", "protobuf", "pyyaml", "pybind11"],
     cmdclass={"build_native": BuildNative},
     entry_points={
         "console_scripts": [
