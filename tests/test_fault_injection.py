@@ -122,3 +122,39 @@ def test_patch_conflict_storm(stack):
     # lifecycle) succeeds
     resp = alloc.allocate(_request(8))
     assert resp.container_responses[0].envs[consts.ENV_RESOURCE_INDEX] == "0"
+
+
+def test_concurrent_allocates_informer_mode(stack):
+    """16 same-size pods, 16 concurrent Allocates against the watch-backed
+    store (the production informer path): every pod assigned exactly once,
+    no poisoned responses, no double assignment."""
+    import threading
+
+    server, informer, alloc = stack
+    for i in range(16):
+        server.store.add_pod(
+            make_pod(f"p{i}", node=NODE, mem=4, gpu_idx=i % 2, assume_time_ns=i)
+        )
+    assert wait_for(lambda: len(informer.pods()) == 16)
+    results = []
+    lock = threading.Lock()
+
+    def run():
+        resp = alloc.allocate(_request(4))
+        with lock:
+            results.append(
+                resp.container_responses[0].envs[consts.ENV_RESOURCE_INDEX]
+            )
+
+    threads = [threading.Thread(target=run) for _ in range(16)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert "-1" not in results
+    assigned = [
+        p
+        for p in server.store.pods.values()
+        if p["metadata"]["annotations"][consts.ENV_ASSIGNED_FLAG] == "true"
+    ]
+    assert len(assigned) == 16
