@@ -65,8 +65,16 @@ def main(argv=None) -> int:
 
     kube = RestKubeClient(base_url=args.api_url) if args.api_url else RestKubeClient()
     extender = GPUShareExtender(kube, resync_interval=args.resync_interval)
-    n = discover_nodes(kube, extender)
-    extender.resync()
+    # initial discovery retries: the apiserver may lag the daemon at boot
+    n = 0
+    for attempt in range(30):
+        try:
+            n = discover_nodes(kube, extender)
+            extender.resync()
+            break
+        except Exception as e:  # noqa: BLE001
+            log.warning("initial node discovery failed (%s); retrying", e)
+            time.sleep(min(2.0 * (attempt + 1), 10.0))
 
     informer = None
     if not args.no_watch:
@@ -80,8 +88,11 @@ def main(argv=None) -> int:
     try:
         while True:
             time.sleep(args.resync_interval)
-            discover_nodes(kube, extender)
-            extender.resync()
+            try:
+                discover_nodes(kube, extender)
+                extender.resync()
+            except Exception as e:  # noqa: BLE001 — an apiserver blip must
+                log.warning("resync failed: %s", e)   # not kill the daemon
     except KeyboardInterrupt:
         if informer is not None:
             informer.stop()
