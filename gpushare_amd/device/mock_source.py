@@ -2,9 +2,15 @@
 
 The reference has no mock backend at all (its tests cannot run without a
 node); this one makes every control-plane code path testable on
-CPU.  Spec grammar: ``"<count>x<mem><GiB|MiB>"`` — e.g. ``"1x8GiB"`` (config
-1), ``"8x288GiB"`` (an 8×MI355X node).  Health events can be injected
-programmatically for watcher tests.
+CPU.  Spec grammar:
+
+- ``"<count>x<mem><GiB|MiB>"`` — homogeneous, e.g. ``"1x8GiB"`` (config 1),
+  ``"8x288GiB"`` (an 8×MI355X node);
+- ``"<m1>+<m2>+...<GiB|MiB>"`` — heterogeneous per-GPU capacities, e.g.
+  ``"288+288+96GiB"`` (the reference assumes homogeneous nodes,
+  nvidia.go:70-72; this build does not).
+
+Health events can be injected programmatically for watcher tests.
 """
 
 from __future__ import annotations
@@ -16,6 +22,7 @@ from typing import Iterable, Optional
 from . import HealthEvent, PhysicalGPU
 
 _SPEC_RE = re.compile(r"^(\d+)x(\d+)(GiB|MiB)$")
+_HETERO_RE = re.compile(r"^(\d+(?:\+\d+)+)(GiB|MiB)$")
 
 
 class MockSource:
@@ -25,18 +32,26 @@ class MockSource:
 
     @classmethod
     def from_spec(cls, spec: str) -> "MockSource":
-        m = _SPEC_RE.match(spec.strip())
-        if not m:
-            raise ValueError(
-                f"bad GPUSHARE_MOCK_SPEC {spec!r}; expected e.g. '1x8GiB'"
-            )
-        count, mem, unit = int(m.group(1)), int(m.group(2)), m.group(3)
+        spec = spec.strip()
+        m = _SPEC_RE.match(spec)
+        if m:
+            count, mem, unit = int(m.group(1)), int(m.group(2)), m.group(3)
+            mems = [mem] * count
+        else:
+            h = _HETERO_RE.match(spec)
+            if not h:
+                raise ValueError(
+                    f"bad GPUSHARE_MOCK_SPEC {spec!r}; expected e.g. "
+                    f"'1x8GiB' or '288+288+96GiB'"
+                )
+            mems, unit = [int(x) for x in h.group(1).split("+")], h.group(2)
+            count = len(mems)
         shift = 30 if unit == "GiB" else 20
         gpus = [
             PhysicalGPU(
                 index=i,
                 uuid=f"mock-{i:02d}",
-                memory_bytes=mem << shift,
+                memory_bytes=mems[i] << shift,
                 render_path=f"/dev/dri/renderD{128 + i}",
                 card_path=f"/dev/dri/card{i}",
                 bdf=f"0000:{0x10 + i:02x}:00.0",

@@ -121,3 +121,13 @@ def test_kfd_resolve(fake_kfd, tmp_path):
     assert t.xgmi_peer_gpu_ids == [5678]
     assert t.vram_bytes == 8 << 30
     assert t.gfx_target_version == 90500
+
+
+def test_mock_heterogeneous_spec():
+    """'288+288+96GiB' — per-GPU capacities differ (the reference assumes a
+    homogeneous node, nvidia.go:70-72; we advertise real per-GPU sizes)."""
+    from gpushare_amd.device.mock_source import MockSource
+
+    gpus = MockSource.from_spec("288+288+96GiB").devices()
+    assert [g.mem_units("GiB") for g in gpus] == [288, 288, 96]
+    assert gpus[2].xgmi_peers == (0, 1)

@@ -319,3 +319,38 @@ class TestPlacementRecordFallback:
         ext.resync()
         assert ext._placements == {}
         assert ext.state.packing()["allocated_units"] == 0
+
+
+class TestHeterogeneousNode:
+    def test_topology_drives_per_gpu_capacities(self):
+        """[288,288,96] node: the extender must reject a 100 GiB pod from
+        landing on the 96 GiB GPU and place it on a 288 GiB one — possible
+        only because the plugin publishes REAL per-GPU capacities (the
+        reference derives total/count and would misplace)."""
+        gpus = MockSource.from_spec("288+288+96GiB").devices()
+        kube = FakeKubeClient(node_name=NODE)
+        pm = PodManager(kube, NODE, kubelet_client=None, query_kubelet=False)
+        pm.patch_topology(gpus)
+        kube.patch_node_status(
+            NODE,
+            {"status": {"allocatable": {
+                consts.RESOURCE_COUNT: "3",
+                consts.RESOURCE_NAME: str(288 + 288 + 96),
+            }}},
+        )
+        from gpushare_amd.extender.__main__ import discover_nodes
+
+        ext = GPUShareExtender(kube, resync_interval=3600)
+        assert discover_nodes(kube, ext) == 1
+        assert ext.state.nodes[NODE].per_gpu_units == [288, 288, 96]
+
+        # fill both 288s so only the 96 has room; a 100 GiB pod must span
+        for name, mem in (("a", 288), ("b", 288)):
+            pod = make_pod(name, node=NODE, mem=mem)
+            del pod["metadata"]["annotations"]
+            kube.add_pod(pod)
+            assert ext.assume(pod, NODE) is not None
+        pod = make_pod("c", node=NODE, mem=100)
+        del pod["metadata"]["annotations"]
+        kube.add_pod(pod)
+        assert ext.assume(pod, NODE) is None  # 96 < 100 and others full
