@@ -226,3 +226,39 @@ def test_health_monitor_deep_probe_on_gpu(source, tmp_path):
         assert mon._probe_failed == set()
     finally:
         mon.stop()
+
+
+@pytest.mark.gpu
+def test_topology_annotation_from_real_device(source):
+    """patch_topology publishes the real per-GPU capacity + xGMI list from
+    KFD; the extender's discover path must accept it round-trip."""
+    import json
+
+    from gpushare_amd import consts
+    from gpushare_amd.cluster.kubeclient import FakeKubeClient
+    from gpushare_amd.cluster.podmanager import PodManager
+    from gpushare_amd.extender.__main__ import discover_nodes
+    from gpushare_amd.extender.core import GPUShareExtender
+
+    gpus = source.devices()
+    kube = FakeKubeClient(node_name="gpu-node")
+    pm = PodManager(kube, "gpu-node", kubelet_client=None, query_kubelet=False)
+    pm.patch_topology(gpus)
+    topo = json.loads(
+        kube.get_node("gpu-node")["metadata"]["annotations"][
+            consts.ANN_NODE_TOPOLOGY
+        ]
+    )
+    assert topo["per_gpu_units"][0] >= 200  # MI355X: ~288 GiB advertised
+    assert len(topo["xgmi"]) == len(gpus)
+
+    kube.patch_node_status(
+        "gpu-node",
+        {"status": {"allocatable": {
+            consts.RESOURCE_COUNT: str(len(gpus)),
+            consts.RESOURCE_NAME: str(sum(topo["per_gpu_units"])),
+        }}},
+    )
+    ext = GPUShareExtender(kube, resync_interval=3600)
+    assert discover_nodes(kube, ext) == 1
+    assert ext.state.nodes["gpu-node"].per_gpu_units == topo["per_gpu_units"]
