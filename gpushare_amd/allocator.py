@@ -91,6 +91,7 @@ class Allocator:
         disable_isolation: bool = False,
         inject_devices: bool = True,
         event_recorder=None,
+        memguard_path: str = "",
     ):
         self.gpus = {g.index: g for g in gpus}
         self.pods = pod_manager
@@ -98,6 +99,10 @@ class Allocator:
         self.disable_isolation = disable_isolation
         self.inject_devices = inject_devices
         self.events = event_recorder
+        # host path of libgpushare_memguard.so; when set (and isolation is
+        # not disabled) Allocate injects the LD_PRELOAD VRAM budget enforcer
+        # — the MI355X answer to the reference's closed-source cGPU module
+        self.memguard_path = memguard_path
         self.stats = AllocateStats()
         # Matching runs under a short in-memory critical section; the
         # ASSIGNED patch happens OUTSIDE it.  A matched pod is "claimed"
@@ -286,6 +291,16 @@ class Allocator:
             c.envs[consts.ENV_RESOURCE_BY_DEV] = str(dev_units)
             if self.disable_isolation:
                 c.envs[consts.ENV_CGPU_DISABLE] = "true"
+            elif self.memguard_path:
+                shift = 30 if self.unit == consts.GIB else 20
+                c.envs[consts.ENV_MEMGUARD_LIMIT] = str(
+                    len(cr.devicesIDs) << shift
+                )
+                c.envs["LD_PRELOAD"] = consts.MEMGUARD_CONTAINER_PATH
+                m = c.mounts.add()
+                m.container_path = consts.MEMGUARD_CONTAINER_PATH
+                m.host_path = self.memguard_path
+                m.read_only = True
             if self.inject_devices:
                 paths = [consts.DEV_KFD]
                 for g in gpus:

@@ -61,6 +61,12 @@ def parse_args(argv=None):
     p.add_argument("--no-informer", action="store_true",
                    help="disable the pod watch informer; every Allocate "
                         "lists pods remotely (reference behavior)")
+    p.add_argument("--memguard-dir", default="", metavar="HOSTDIR",
+                   help="enable per-container VRAM budget enforcement: "
+                        "copy libgpushare_memguard.so into this hostPath "
+                        "dir and inject it (LD_PRELOAD) into every "
+                        "allocated container; empty = advisory isolation "
+                        "only (reference behavior without cGPU)")
     p.add_argument("--metrics-port", type=int, default=0,
                    help="serve Prometheus /metrics on this port (0 = off)")
     p.add_argument("-v", "--verbose", action="count", default=0)
@@ -87,6 +93,27 @@ def main(argv=None) -> int:
 
         metrics.serve(args.metrics_port)
 
+    memguard_path = ""
+    if args.memguard_dir:
+        # stage the packaged enforcer onto the hostPath the kubelet will
+        # bind-mount into allocated containers
+        import shutil
+
+        import gpushare_amd
+
+        src = os.path.join(
+            os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+        )
+        if not os.path.exists(src):
+            log.error("--memguard-dir set but %s is not built", src)
+            return 2
+        os.makedirs(args.memguard_dir, exist_ok=True)
+        memguard_path = os.path.join(
+            args.memguard_dir, "libgpushare_memguard.so"
+        )
+        shutil.copy2(src, memguard_path)
+        log.info("memguard staged at %s", memguard_path)
+
     source = create_source(args.mock_spec)
     kube = RestKubeClient()
     kubelet = None
@@ -112,6 +139,7 @@ def main(argv=None) -> int:
             cache_ttl=args.cache_ttl,
             inject_devices=not args.no_inject,
             use_informer=not args.no_informer,
+            memguard_path=memguard_path,
         ),
     )
     mgr.install_signal_handlers()

@@ -71,6 +71,13 @@ ENV_ROCR_VISIBLE = "ROCR_VISIBLE_DEVICES"
 DEV_KFD = "/dev/kfd"
 DEV_DRI_DIR = "/dev/dri"
 
+# LD_PRELOAD VRAM budget enforcer (native/memguard.cpp) — the MI355X-native
+# replacement for the reference's closed-source cGPU isolation module.
+# Allocate() mounts the library read-only at MEMGUARD_CONTAINER_PATH and
+# sets ENV_MEMGUARD_LIMIT to the container's gpu-mem share in bytes.
+ENV_MEMGUARD_LIMIT = "GPUSHARE_MEM_LIMIT_BYTES"
+MEMGUARD_CONTAINER_PATH = "/usr/local/lib/gpushare/libgpushare_memguard.so"
+
 # Node label opting a node out of kernel-level isolation (analogue of the
 # reference's cgpu.disable.isolation toggle, podmanager.go:59-72).
 LABEL_DISABLE_ISOLATION = "cgpu.disable.isolation"

@@ -48,6 +48,8 @@ class ManagerOptions:
     coredump_dir: str = "/etc/kubernetes"
     use_informer: bool = True    # watch-based pending-pod cache (informer.py);
                                  # falls back to LIST paths while unsynced
+    memguard_path: str = ""      # host path of libgpushare_memguard.so;
+                                 # "" disables VRAM budget enforcement
 
 
 class SharedGPUManager:
@@ -101,6 +103,7 @@ class SharedGPUManager:
             disable_isolation=pm.isolation_disabled(),
             inject_devices=self.opt.inject_devices,
             event_recorder=EventRecorder(self.kube, self.node_name),
+            memguard_path=self.opt.memguard_path,
         )
         return GPUSharePlugin(
             gpus,

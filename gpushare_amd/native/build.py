@@ -6,6 +6,8 @@ repo snapshot:
   gpushare_amd._amdsmi  — C++  dlopen shim over libamd_smi.so   (g++)
   gpushare_amd._devlist — C++  ListAndWatchResponse pre-encoder (g++)
   gpushare_amd._canary  — HIP  gfx950 health-probe kernels      (hipcc)
+  libgpushare_memguard.so — C++ LD_PRELOAD VRAM budget enforcer (g++;
+                            plain shared lib, not a Python extension)
 
 Run:  python -m gpushare_amd.native.build  [--force]
 """
@@ -64,6 +66,14 @@ def build(force: bool = False, verbose: bool = True) -> list[Path]:
             HERE / "canary.hip",
             PKG / f"_canary{EXT_SUFFIX}",
             [f"--offload-arch={GPU_ARCH}", "-O3"],
+        ),
+        (
+            "g++",
+            HERE / "memguard.cpp",
+            PKG / "libgpushare_memguard.so",
+            # visibility=default overrides _COMMON's hidden: the interposed
+            # hip* symbols must be exported for LD_PRELOAD to work
+            ["-fvisibility=default", "-ldl"],
         ),
     ]
     for compiler, src, out, extra in targets:

@@ -176,3 +176,25 @@ def test_stats_percentiles_populated():
     assert snap["count"] == 5
     assert snap["failures"] == 0
     assert snap["p50_ms"] > 0
+
+
+def test_memguard_injection():
+    """--memguard-dir mode: Allocate mounts the enforcer read-only and sets
+    LD_PRELOAD + the per-container byte budget; disable_isolation wins."""
+    kube = FakeKubeClient(node_name="node-a")
+    pm = PodManager(kube, "node-a", kubelet_client=kube.as_kubelet(),
+                    cache_ttl=0.0, kubelet_retries=0, apiserver_retries=0)
+    gpus = MockSource.from_spec("1x16GiB").devices()
+    alloc = Allocator(gpus, pm, memguard_path="/var/lib/gpushare/libgpushare_memguard.so")
+    resp = alloc.allocate(_request([4]))
+    c = resp.container_responses[0]
+    assert c.envs[consts.ENV_MEMGUARD_LIMIT] == str(4 << 30)
+    assert c.envs["LD_PRELOAD"] == consts.MEMGUARD_CONTAINER_PATH
+    mounts = [(m.host_path, m.container_path, m.read_only) for m in c.mounts]
+    assert mounts == [("/var/lib/gpushare/libgpushare_memguard.so",
+                       consts.MEMGUARD_CONTAINER_PATH, True)]
+
+    alloc2 = Allocator(gpus, pm, memguard_path="/x.so", disable_isolation=True)
+    c2 = alloc2.allocate(_request([4])).container_responses[0]
+    assert consts.ENV_MEMGUARD_LIMIT not in c2.envs
+    assert len(c2.mounts) == 0

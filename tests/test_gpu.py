@@ -262,3 +262,46 @@ def test_topology_annotation_from_real_device(source):
     ext = GPUShareExtender(kube, resync_interval=3600)
     assert discover_nodes(kube, ext) == 1
     assert ext.state.nodes["gpu-node"].per_gpu_units == topo["per_gpu_units"]
+
+
+@pytest.mark.gpu
+def test_memguard_enforces_vram_budget():
+    """LD_PRELOAD enforcement on a real MI355X: a 2 GiB budget lets a
+    1 GiB tensor through, clamps hipMemGetInfo to the budget, and turns
+    an over-budget allocation into an ordinary torch OOM."""
+    import os
+    import subprocess
+    import sys
+
+    import gpushare_amd
+
+    lib = os.path.join(
+        os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+    )
+    assert os.path.exists(lib), "memguard not built"
+    script = r"""
+import torch
+assert torch.cuda.is_available()
+free, total = torch.cuda.mem_get_info()
+assert total <= (2 << 30), f"hipMemGetInfo not clamped: total={total}"
+a = torch.empty(1 << 30, dtype=torch.uint8, device="cuda:0")  # 1 GiB: fits
+try:
+    b = torch.empty(2 << 30, dtype=torch.uint8, device="cuda:0")
+except torch.OutOfMemoryError:
+    print("MEMGUARD_OK")
+else:
+    print("MEMGUARD_FAIL no OOM")
+"""
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = lib
+    env["GPUSHARE_MEM_LIMIT_BYTES"] = str(2 << 30)
+    out = subprocess.run(
+        [sys.executable, "-c", script],
+        env=env,
+        capture_output=True,
+        text=True,
+        timeout=180,
+    )
+    assert "MEMGUARD_OK" in out.stdout, (
+        f"stdout={out.stdout!r} stderr={out.stderr[-2000:]!r}"
+    )
