@@ -60,10 +60,30 @@ void init_limit() {
     });
 }
 
+void* hip_handle() {
+    // RTLD_NEXT only scans the GLOBAL scope after this library.  PyTorch
+    // loads libamdhip64 as a dependency of an RTLD_LOCAL extension module,
+    // so the runtime is absent from the global scope — grab a handle to
+    // the already-loaded copy by soname instead (RTLD_NOLOAD never maps a
+    // second copy; the final plain dlopen covers a not-yet-loaded runtime
+    // and resolves to the same file the app will get).
+    static void* h = [] {
+        for (const char* name :
+             {"libamdhip64.so.7", "libamdhip64.so.6", "libamdhip64.so"}) {
+            void* p = dlopen(name, RTLD_LAZY | RTLD_NOLOAD);
+            if (p) return p;
+        }
+        return dlopen("libamdhip64.so", RTLD_LAZY);
+    }();
+    return h;
+}
+
 template <typename Fn>
 Fn real(const char* name) {
-    // resolved from the app's own HIP runtime; null if no HIP loaded
-    return reinterpret_cast<Fn>(dlsym(RTLD_NEXT, name));
+    void* sym = dlsym(RTLD_NEXT, name);
+    if (sym == nullptr && hip_handle() != nullptr)
+        sym = dlsym(hip_handle(), name);
+    return reinterpret_cast<Fn>(sym);
 }
 
 bool reserve(size_t size) {
