@@ -124,6 +124,20 @@ class AmdSmiSource:
     def devices(self) -> list[PhysicalGPU]:
         return list(self._gpus)
 
+    def vram_usage(self) -> dict[int, int]:
+        """Live per-GPU VRAM used bytes (observability; the scheduling
+        currency stays the *allocated* annotations, not live usage)."""
+        out: dict[int, int] = {}
+        for g in self._gpus:
+            try:
+                info = self._smi.device_info(g.index)
+            except RuntimeError:
+                continue
+            used = info.get("vram_used_bytes")
+            if used is not None:
+                out[g.index] = int(used)
+        return out
+
     # ------------------------------------------------------------------ #
     def watch_health(self, stop_event) -> Iterable[HealthEvent]:
         smi = self._smi

@@ -83,7 +83,14 @@ class HealthMonitor:
         except ImportError as e:
             log.error("deep probe requested but _canary not built: %s", e)
             return
+        from . import metrics
+
         while not self._stop.wait(self.deep_probe_interval):
+            if hasattr(self.source, "vram_usage"):
+                try:
+                    metrics.observe_vram_usage(self.source.vram_usage())
+                except Exception as e:  # noqa: BLE001
+                    log.warning("vram usage poll failed: %s", e)
             for gpu in self.plugin.gpus:
                 if self._stop.is_set():
                     return

@@ -65,6 +65,9 @@ if AVAILABLE:
     INFORMER_SYNCED = Gauge(
         "gpushare_informer_synced", "Pod informer synced (1) / degraded (0)"
     )
+    VRAM_USED = Gauge(
+        "gpushare_vram_used_bytes", "Live VRAM usage per GPU", ["gpu"]
+    )
 
 
 def observe_allocate(total_s: float, list_s: float, patch_s: float, ok: bool) -> None:
@@ -91,6 +94,13 @@ def observe_inventory(n_devices: int) -> None:
 def observe_law_send() -> None:
     if AVAILABLE:
         LISTANDWATCH_SENDS.inc()
+
+
+def observe_vram_usage(usage: dict) -> None:
+    if not AVAILABLE:
+        return
+    for gpu_idx, used in usage.items():
+        VRAM_USED.labels(str(gpu_idx)).set(used)
 
 
 def observe_informer_event() -> None:
