@@ -134,7 +134,9 @@ class Allocator:
                 if not gpus:
                     self._unclaim(uid)
                     return self._err_response(request, req_units)
-                resp = self._build_response(request, req_units, gpus)
+                resp = self._build_response(
+                    request, req_units, gpus, pod_uid=uid
+                )
                 tp = time.perf_counter()
                 patched = self.pods.mark_assigned(pod)
                 patch_t = time.perf_counter() - tp
@@ -267,7 +269,11 @@ class Allocator:
 
     # ------------------------------------------------------------------ #
     def _build_response(
-        self, request, req_units: int, gpus: list[PhysicalGPU]
+        self,
+        request,
+        req_units: int,
+        gpus: list[PhysicalGPU],
+        pod_uid: str = "",
     ) -> "api.AllocateResponse":
         """Container envs + AMD device nodes for the bound GPU(s).  For a
         multi-GPU placement every container sees the whole set (ROCr
@@ -297,6 +303,9 @@ class Allocator:
                     len(cr.devicesIDs) << shift
                 )
                 c.envs["LD_PRELOAD"] = consts.MEMGUARD_CONTAINER_PATH
+                if pod_uid:
+                    # scopes the shared budget counter under /dev/shm
+                    c.envs[consts.ENV_MEMGUARD_POD_UID] = pod_uid
                 m = c.mounts.add()
                 m.container_path = consts.MEMGUARD_CONTAINER_PATH
                 m.host_path = self.memguard_path

@@ -76,6 +76,7 @@ DEV_DRI_DIR = "/dev/dri"
 # Allocate() mounts the library read-only at MEMGUARD_CONTAINER_PATH and
 # sets ENV_MEMGUARD_LIMIT to the container's gpu-mem share in bytes.
 ENV_MEMGUARD_LIMIT = "GPUSHARE_MEM_LIMIT_BYTES"
+ENV_MEMGUARD_POD_UID = "GPUSHARE_POD_UID"
 MEMGUARD_CONTAINER_PATH = "/usr/local/lib/gpushare/libgpushare_memguard.so"
 
 # Node label opting a node out of kernel-level isolation (analogue of the

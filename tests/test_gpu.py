@@ -305,3 +305,72 @@ else:
     assert "MEMGUARD_OK" in out.stdout, (
         f"stdout={out.stdout!r} stderr={out.stderr[-2000:]!r}"
     )
+
+
+@pytest.mark.gpu
+def test_memguard_budget_shared_across_processes():
+    """Two processes under ONE pod budget (shm counter): while A holds
+    1.5 GiB of a 2 GiB budget, B's 1 GiB allocation OOMs; after A exits
+    (budget repaid), B's retry succeeds."""
+    import os
+    import subprocess
+    import sys
+    import time
+    import uuid as uuid_mod
+
+    import gpushare_amd
+
+    lib = os.path.join(
+        os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+    )
+    pod_uid = f"test-{uuid_mod.uuid4().hex[:12]}"
+    shm_path = f"/dev/shm/gpushare.memguard.{pod_uid}"
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = lib
+    env["GPUSHARE_MEM_LIMIT_BYTES"] = str(2 << 30)
+    env["GPUSHARE_POD_UID"] = pod_uid
+
+    holder = subprocess.Popen(
+        [sys.executable, "-c", (
+            "import torch, time, sys;"
+            "a=torch.empty(3<<29, dtype=torch.uint8, device='cuda:0');"
+            "print('HELD', flush=True); time.sleep(60)"
+        )],
+        env=env, stdout=subprocess.PIPE, text=True,
+    )
+    try:
+        line = holder.stdout.readline()
+        assert "HELD" in line, f"holder failed: {line}"
+        probe = (
+            "import torch;\n"
+            "try:\n"
+            "    b=torch.empty(1<<30, dtype=torch.uint8, device='cuda:0')\n"
+            "    print('ALLOC_OK', flush=True)\n"
+            "except torch.OutOfMemoryError:\n"
+            "    print('ALLOC_OOM', flush=True)\n"
+        )
+        out = subprocess.run(
+            [sys.executable, "-c", probe], env=env,
+            capture_output=True, text=True, timeout=180,
+        )
+        assert "ALLOC_OOM" in out.stdout, (
+            f"expected shared-budget OOM: {out.stdout!r} {out.stderr[-800:]!r}"
+        )
+    finally:
+        holder.terminate()
+        holder.wait(timeout=30)
+    # holder's exit repaid its reservation -> the same alloc now fits
+    deadline = time.monotonic() + 30
+    ok = False
+    while time.monotonic() < deadline:
+        out = subprocess.run(
+            [sys.executable, "-c", probe], env=env,
+            capture_output=True, text=True, timeout=180,
+        )
+        if "ALLOC_OK" in out.stdout:
+            ok = True
+            break
+        time.sleep(2)
+    if os.path.exists(shm_path):
+        os.unlink(shm_path)
+    assert ok, "budget was not repaid after holder exit"
