@@ -131,3 +131,42 @@ def test_mock_heterogeneous_spec():
     gpus = MockSource.from_spec("288+288+96GiB").devices()
     assert [g.mem_units("GiB") for g in gpus] == [288, 288, 96]
     assert gpus[2].xgmi_peers == (0, 1)
+
+
+def test_memguard_lib_loads_and_reads_env(tmp_path):
+    """CPU-side smoke of libgpushare_memguard.so: the lib must dlopen
+    anywhere (links only libdl/libc), parse the env budget, and map its
+    pod-scoped slot table (no HIP runtime involved)."""
+    import ctypes
+    import os
+    import subprocess
+    import sys
+
+    import gpushare_amd
+
+    lib = os.path.join(
+        os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+    )
+    if not os.path.exists(lib):
+        import pytest
+
+        pytest.skip("memguard not built")
+    code = (
+        "import ctypes, os\n"
+        f"l = ctypes.CDLL({lib!r})\n"
+        "l.gpushare_memguard_limit.restype = ctypes.c_int64\n"
+        "l.gpushare_memguard_used.restype = ctypes.c_int64\n"
+        "assert l.gpushare_memguard_limit() == 1 << 30\n"
+        "assert l.gpushare_memguard_used() == 0\n"
+        "print('CPU_MEMGUARD_OK')\n"
+    )
+    env = dict(os.environ)
+    env["GPUSHARE_MEM_LIMIT_BYTES"] = str(1 << 30)
+    env["GPUSHARE_POD_UID"] = f"cputest-{os.getpid()}"
+    out = subprocess.run(
+        [sys.executable, "-c", code], env=env, capture_output=True, text=True
+    )
+    shm = f"/dev/shm/gpushare.memguard.cputest-{os.getpid()}"
+    if os.path.exists(shm):
+        os.unlink(shm)
+    assert "CPU_MEMGUARD_OK" in out.stdout, out.stderr
