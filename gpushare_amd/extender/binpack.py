@@ -124,6 +124,40 @@ class BinpackState:
                     out.append(name)
             return out
 
+    def score_nodes(
+        self, request: int, candidates: list[str], max_score: int = 10
+    ) -> dict[str, int]:
+        """Binpack node scoring for the scheduler `prioritize` webhook:
+        higher for the node whose placement leaves the least slack — pods
+        funnel onto the fullest feasible node, keeping whole GPUs (and
+        whole nodes) free for large/multi-GPU pods."""
+        with self._lock:
+            slacks: dict[str, int] = {}
+            for name in candidates:
+                st = self.nodes.get(name)
+                if st is None:
+                    continue
+                split = st.best_fit_multi(request)
+                if split is None:
+                    slacks[name] = -1  # infeasible: score 0
+                    continue
+                slacks[name] = sum(st.free(i) for i in split) - request
+            feasible = {n: s for n, s in slacks.items() if s >= 0}
+            scores = {n: 0 for n in slacks}  # infeasible: 0
+            if not feasible:
+                return scores
+            best = min(feasible.values())
+            worst = max(feasible.values())
+            for name, slack in feasible.items():
+                if worst == best:
+                    scores[name] = max_score
+                else:
+                    # tightest fit -> max_score, loosest feasible -> 1
+                    scores[name] = 1 + round(
+                        (max_score - 1) * (worst - slack) / (worst - best)
+                    )
+            return scores
+
     def assume(self, node: str, request: int) -> Optional[int]:
         """Reserve `request` units on the best-fit single GPU; returns the
         GPU index (legacy single-GPU path; multi: assume_multi)."""

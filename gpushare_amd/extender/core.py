@@ -209,6 +209,19 @@ class GPUShareExtender:
             return None
         return pod
 
+    def prioritize(self, pod: dict, node_names: list[str]) -> list[dict]:
+        """Webhook `prioritize`: HostPriorityList scoring feasible nodes by
+        binpack tightness (k8s extender max score 10).  The reference
+        extender exposes filter+bind only; without scoring the default
+        spreading policy works against co-location."""
+        request = podutils.gpu_memory_of_pod(pod)
+        if request <= 0:
+            return [{"Host": n, "Score": 0} for n in node_names]
+        scores = self.state.score_nodes(request, node_names)
+        return [
+            {"Host": n, "Score": scores.get(n, 0)} for n in node_names
+        ]
+
     def packing(self) -> dict:
         return self.state.packing()
 

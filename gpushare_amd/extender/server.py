@@ -6,6 +6,7 @@ at the same URL prefix the gpushare scheduler-extender uses, plus a
 ``release`` hook (informer delete event) and a ``packing`` report.
 
   POST /gpushare-scheduler/filter
+  POST /gpushare-scheduler/prioritize   (binpack node scoring)
   POST /gpushare-scheduler/bind
   POST /gpushare-scheduler/release
   GET  /gpushare-scheduler/packing
@@ -66,6 +67,15 @@ class ExtenderServer:
             return 200, json.dumps(
                 {"NodeNames": ok, "FailedNodes": {}, "Error": ""}
             ).encode()
+        if path == "/gpushare-scheduler/prioritize":
+            pod = body.get("Pod") or {}
+            names = body.get("NodeNames") or [
+                n.get("metadata", {}).get("name")
+                for n in (body.get("Nodes") or {}).get("Items", [])
+            ]
+            return 200, json.dumps(
+                ext.prioritize(pod, [n for n in names if n])
+            ).encode()
         if path == "/gpushare-scheduler/bind":
             ns = body.get("PodNamespace", "default")
             name = body.get("PodName", "")
@@ -111,6 +121,12 @@ class ExtenderClient:
             "/gpushare-scheduler/filter",
             {"Pod": pod, "NodeNames": node_names},
         )["NodeNames"]
+
+    def prioritize(self, pod: dict, node_names: list[str]) -> list[dict]:
+        return self._post(
+            "/gpushare-scheduler/prioritize",
+            {"Pod": pod, "NodeNames": node_names},
+        )
 
     def bind(self, namespace: str, name: str, node: str) -> str:
         return self._post(

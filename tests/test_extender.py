@@ -203,3 +203,45 @@ def test_pod_manager_against_http_apiserver():
         client.close()
     finally:
         api.stop()
+
+
+class TestPrioritize:
+    def test_binpack_scoring_prefers_fuller_node(self):
+        """prioritize: the node whose placement leaves less slack scores
+        higher (co-location over spreading); infeasible nodes score 0."""
+        from gpushare_amd.cluster.kubeclient import FakeKubeClient
+        from gpushare_amd.extender.core import GPUShareExtender
+
+        kube = FakeKubeClient(node_name="a")
+        ext = GPUShareExtender(kube, resync_interval=3600)
+        ext.register_node("a", [16, 16])
+        ext.register_node("b", [16, 16])
+        ext.register_node("tiny", [4])
+        # node a partially used -> tighter fit for an 8 GiB pod
+        ext.state.assume("a", 6)
+
+        pod = make_pod("p", mem=8, node="")
+        scores = {
+            e["Host"]: e["Score"]
+            for e in ext.prioritize(pod, ["a", "b", "tiny"])
+        }
+        assert scores["a"] > scores["b"]
+        assert scores["tiny"] == 0       # cannot fit 8 GiB
+        assert 0 <= max(scores.values()) <= 10
+
+    def test_prioritize_over_http(self):
+        from gpushare_amd.cluster.kubeclient import FakeKubeClient
+        from gpushare_amd.extender.core import GPUShareExtender
+        from gpushare_amd.extender.server import ExtenderClient, ExtenderServer
+
+        kube = FakeKubeClient(node_name="a")
+        ext = GPUShareExtender(kube, resync_interval=3600)
+        ext.register_node("a", [16])
+        server = ExtenderServer(ext).start()
+        try:
+            client = ExtenderClient(server.url)
+            out = client.prioritize(make_pod("p", mem=8, node=""), ["a"])
+            assert out == [{"Host": "a", "Score": 10}]
+            client.close()
+        finally:
+            server.stop()
