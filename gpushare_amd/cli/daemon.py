@@ -61,6 +61,10 @@ def parse_args(argv=None):
     p.add_argument("--no-informer", action="store_true",
                    help="disable the pod watch informer; every Allocate "
                         "lists pods remotely (reference behavior)")
+    p.add_argument("--allow-oversize-inventory", action="store_true",
+                   help="serve a ListAndWatch payload above the kubelet's "
+                        "4 MiB gRPC receive default (requires a patched "
+                        "kubelet; MiB grain on 288 GiB GPUs needs this)")
     p.add_argument("--memguard-dir", default="", metavar="HOSTDIR",
                    help="enable per-container VRAM budget enforcement: "
                         "copy libgpushare_memguard.so into this hostPath "
@@ -140,6 +144,7 @@ def main(argv=None) -> int:
             inject_devices=not args.no_inject,
             use_informer=not args.no_informer,
             memguard_path=memguard_path,
+            allow_oversize_inventory=args.allow_oversize_inventory,
         ),
     )
     mgr.install_signal_handlers()

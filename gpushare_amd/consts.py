@@ -25,6 +25,12 @@ SERVER_SOCK_NAME = "amdgpushare.sock"
 # Max length of a Device.ID on the wire (api.proto:84).
 MAX_DEVICE_ID_LEN = 63
 
+# The kubelet's gRPC client receives with the Go default 4 MiB cap; a
+# ListAndWatchResponse above it is silently fatal (RESOURCE_EXHAUSTED on
+# the kubelet side).  At MiB grain one 288 GiB MI355X already encodes to
+# ~8.4 MB — see GPUSharePlugin's inventory guard.
+KUBELET_GRPC_MAX_MSG = 4 << 20
+
 # ---------------------------------------------------------------------------
 # Extended resources (scheduler contract — unchanged from the reference so the
 # gpushare-scheduler-extender works as-is; reference: const.go:11-12)

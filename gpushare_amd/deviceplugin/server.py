@@ -42,6 +42,7 @@ class GPUSharePlugin:
         socket_dir: str = consts.DEVICE_PLUGIN_PATH,
         socket_name: str = consts.SERVER_SOCK_NAME,
         max_workers: int = 64,  # each ListAndWatch stream pins a worker thread
+        allow_oversize_inventory: bool = False,
     ):
         self.gpus = gpus
         self.allocator = allocator
@@ -55,6 +56,21 @@ class GPUSharePlugin:
         self.table = FakeDeviceTable.build(gpus, unit)
         self._codec = make_codec(self.table.ids)
         metrics.observe_inventory(len(self.table))
+        # a payload over the kubelet's 4 MiB gRPC receive default fails
+        # RESOURCE_EXHAUSTED on the kubelet side — at MiB grain one 288 GiB
+        # MI355X is already ~8.4 MB.  Fail loudly at startup instead.
+        self.inventory_bytes = len(self._codec.encode([]))
+        if self.inventory_bytes > consts.KUBELET_GRPC_MAX_MSG:
+            msg = (
+                f"inventory of {len(self.table)} fake devices encodes to "
+                f"{self.inventory_bytes / 1e6:.1f} MB — over the kubelet's "
+                f"4 MiB gRPC receive limit; use GiB granularity on "
+                f"288 GiB-class GPUs (or a kubelet built with a larger "
+                f"limit + allow_oversize_inventory)"
+            )
+            if not allow_oversize_inventory:
+                raise ValueError(msg)
+            log.warning("%s — continuing as requested", msg)
 
         # health state: plugin GPU index -> healthy?
         self._health_lock = threading.Condition()

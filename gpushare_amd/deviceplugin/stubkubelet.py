@@ -36,10 +36,21 @@ log = logging.getLogger(__name__)
 class DevicePluginClient:
     """Kubelet's view of one registered plugin."""
 
-    def __init__(self, socket_path: str, resource_name: str):
+    def __init__(
+        self,
+        socket_path: str,
+        resource_name: str,
+        max_receive_mb: int = 4,   # the REAL kubelet's gRPC default — keep
+                                   # the harness as strict as production
+    ):
         self.socket_path = socket_path
         self.resource_name = resource_name
-        self._channel = grpc.insecure_channel(f"unix://{socket_path}")
+        self._channel = grpc.insecure_channel(
+            f"unix://{socket_path}",
+            options=[
+                ("grpc.max_receive_message_length", max_receive_mb << 20)
+            ],
+        )
         grpc.channel_ready_future(self._channel).result(timeout=10)
 
         self._get_options = self._channel.unary_unary(
@@ -192,7 +203,10 @@ class StubKubelet:
         log.info("stub kubelet serving on %s", self.socket_path)
 
     def wait_for_plugin(
-        self, resource_name: str = consts.RESOURCE_NAME, timeout: float = 10.0
+        self,
+        resource_name: str = consts.RESOURCE_NAME,
+        timeout: float = 10.0,
+        max_receive_mb: int = 0,   # >0: reconnect with a raised gRPC limit
     ) -> DevicePluginClient:
         deadline = time.monotonic() + timeout
         with self._reg_lock:
@@ -201,7 +215,17 @@ class StubKubelet:
                 if remaining <= 0:
                     raise TimeoutError(f"plugin {resource_name} never registered")
                 self._reg_lock.wait(timeout=remaining)
-            return self.plugins[resource_name]
+            client = self.plugins[resource_name]
+        if max_receive_mb > 0:
+            bigger = DevicePluginClient(
+                client.socket_path, resource_name,
+                max_receive_mb=max_receive_mb,
+            )
+            with self._reg_lock:
+                self.plugins[resource_name] = bigger
+            client.close()
+            return bigger
+        return client
 
     def stop(self) -> None:
         for client in self.plugins.values():

@@ -50,6 +50,7 @@ class ManagerOptions:
                                  # falls back to LIST paths while unsynced
     memguard_path: str = ""      # host path of libgpushare_memguard.so;
                                  # "" disables VRAM budget enforcement
+    allow_oversize_inventory: bool = False
 
 
 class SharedGPUManager:
@@ -110,6 +111,7 @@ class SharedGPUManager:
             allocator,
             unit=self.opt.memory_unit,
             socket_dir=self.opt.socket_dir,
+            allow_oversize_inventory=self.opt.allow_oversize_inventory,
         )
 
     def _start_plugin(self) -> None:

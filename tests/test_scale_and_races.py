@@ -197,3 +197,28 @@ def test_kubelet_rejects_wrong_api_version(tmp_socket_dir):
             assert err.value.code() == grpc.StatusCode.INVALID_ARGUMENT
     finally:
         kubelet.stop()
+
+
+def test_oversize_inventory_refused():
+    """MiB grain on a 288 GiB GPU encodes past the kubelet's 4 MiB gRPC
+    receive default — a stock kubelet would drop the stream with
+    RESOURCE_EXHAUSTED.  The plugin must refuse loudly at startup instead
+    (override: allow_oversize_inventory for patched kubelets)."""
+    import tempfile
+
+    from gpushare_amd.allocator import Allocator
+    from gpushare_amd.cluster.kubeclient import FakeKubeClient
+    from gpushare_amd.cluster.podmanager import PodManager
+    from gpushare_amd.device.mock_source import MockSource
+    from gpushare_amd.deviceplugin.server import GPUSharePlugin
+
+    gpus = MockSource.from_spec("1x288GiB").devices()
+    kube = FakeKubeClient("node-a")
+    pm = PodManager(kube, "node-a", kubelet_client=None, query_kubelet=False)
+    with tempfile.TemporaryDirectory() as d:
+        with pytest.raises(ValueError, match="4 MiB gRPC receive limit"):
+            GPUSharePlugin(gpus, Allocator(gpus, pm), unit=consts.MIB,
+                           socket_dir=d)
+        p = GPUSharePlugin(gpus, Allocator(gpus, pm), unit=consts.MIB,
+                           socket_dir=d, allow_oversize_inventory=True)
+        assert p.inventory_bytes > consts.KUBELET_GRPC_MAX_MSG
