@@ -115,6 +115,10 @@ class Allocator:
         self._lock = threading.Lock()
         self._claims: dict[str, float] = {}
         self.claim_ttl = 30.0
+        # after a successful ASSIGNED patch the claim only needs to
+        # outlive watch propagation — shrinking it keeps the table small
+        self.claim_grace = 2.0
+        self._last_claim_prune = 0.0
 
     # ------------------------------------------------------------------ #
     def allocate(self, request) -> "api.AllocateResponse":
@@ -143,6 +147,10 @@ class Allocator:
                 if not patched:
                     self._unclaim(uid)
                     return self._err_response(request, req_units)
+                with self._lock:  # durable now: keep only a short grace
+                    self._claims[uid] = (
+                        time.monotonic() + self.claim_grace
+                    )
                 ok = True
                 return resp
             if len(self.gpus) == 1:
@@ -223,10 +231,17 @@ class Allocator:
                 if self._claims.get(uid, 0.0) > now:
                     continue  # being handled by a concurrent Allocate
                 self._claims[uid] = now + self.claim_ttl
-                if len(self._claims) > 10_000:
+                # amortized prune: at most once per second, never on the
+                # per-call path (an O(n) rebuild per Allocate was the p50
+                # driver in long soaks)
+                if (
+                    len(self._claims) > 10_000
+                    and now - self._last_claim_prune > 1.0
+                ):
                     self._claims = {
                         u: t for u, t in self._claims.items() if t > now
                     }
+                    self._last_claim_prune = now
                 return pod
         return None
 

@@ -39,7 +39,8 @@ class GPUShareExtender:
         # event may both fire for one pod — releasing twice would free
         # units another pod holds
         self._released: dict[str, float] = {}
-        self.released_ttl = 60.0
+        self.released_ttl = 10.0  # covers webhook+watch double-fire, keeps the table small
+        self._last_released_prune = 0.0
         # placement record from assume time: release() falls back to it
         # when the caller's pod copy lacks the allocation-map annotation
         # (e.g. a delete-event stub); pruned on resync against live pods
@@ -180,10 +181,15 @@ class GPUShareExtender:
                 if self._released.get(uid, 0.0) > now:
                     return
                 self._released[uid] = now + self.released_ttl
-                if len(self._released) > 10_000:
+                # amortized: never an O(n) rebuild per release
+                if (
+                    len(self._released) > 10_000
+                    and now - self._last_released_prune > 1.0
+                ):
                     self._released = {
                         u: t for u, t in self._released.items() if t > now
                     }
+                    self._last_released_prune = now
         key = (podutils.pod_namespace(pod), podutils.pod_name(pod))
         with self._lock:
             recorded = self._placements.pop(key, None)
