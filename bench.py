@@ -365,8 +365,11 @@ def main():
         packing_peak = max(
             (p for g in gathered for p in g["packing"]), default=0.0
         )
+        import resource
+
         server_stats = plugin.allocator.stats.snapshot()
         pods_per_sec = total_alloc / elapsed_max if elapsed_max > 0 else 0.0
+        rss_mb = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss / 1024
 
         result = {
             "metric": "pods/sec scheduled (gpu-mem share Allocate pipeline)",
@@ -402,6 +405,7 @@ def main():
                 "packing_pct_peak": round(packing_peak, 2),
                 "server_allocate_p50_ms": round(server_stats["p50_ms"], 3),
                 "server_allocate_p99_ms": round(server_stats["p99_ms"], 3),
+                "plugin_rank_rss_mb": round(rss_mb, 1),
             },
         }
         print(json.dumps(result), flush=True)
