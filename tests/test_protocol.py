@@ -103,3 +103,76 @@ def test_codec_scale_2304_devices():
     assert payload == _protobuf_encode(ids, set())
     parsed = api.ListAndWatchResponse.FromString(payload)
     assert len(parsed.devices) == 2304
+
+
+class TestFastHTTPRobustness:
+    def test_garbage_request_drops_connection_not_server(self):
+        """A malformed request line closes that connection; the server
+        keeps serving other clients."""
+        import socket
+
+        from gpushare_amd.cluster.fasthttp import FastHTTPServer
+
+        srv = FastHTTPServer(lambda m, p, b: (200, b'{"ok":true}')).start()
+        try:
+            bad = socket.create_connection(("127.0.0.1", srv.port))
+            bad.sendall(b"NOT-HTTP\r\n\r\n")
+            bad.settimeout(2)
+            assert bad.recv(128) == b""  # dropped
+            bad.close()
+
+            from gpushare_amd.cluster.httpconn import HttpSession
+
+            s = HttpSession(f"http://127.0.0.1:{srv.port}")
+            status, body = s.request("GET", "/x")
+            assert status == 200 and body == b'{"ok":true}'
+            s.close()
+        finally:
+            srv.stop()
+
+    def test_handler_exception_returns_500_keeps_connection(self):
+        from gpushare_amd.cluster.fasthttp import FastHTTPServer
+        from gpushare_amd.cluster.httpconn import HttpSession
+
+        calls = []
+
+        def handler(m, p, b):
+            calls.append(p)
+            if p == "/boom":
+                raise RuntimeError("kaboom")
+            return 200, b"{}"
+
+        srv = FastHTTPServer(handler).start()
+        try:
+            s = HttpSession(f"http://127.0.0.1:{srv.port}")
+            status, body = s.request("GET", "/boom")
+            assert status == 500
+            # same keep-alive connection still serves
+            status, _ = s.request("GET", "/ok")
+            assert status == 200
+            s.close()
+        finally:
+            srv.stop()
+
+    def test_extender_bad_json_returns_500(self):
+        from gpushare_amd.cluster.httpconn import HttpSession
+        from gpushare_amd.cluster.kubeclient import FakeKubeClient
+        from gpushare_amd.extender.core import GPUShareExtender
+        from gpushare_amd.extender.server import ExtenderServer
+
+        ext = GPUShareExtender(FakeKubeClient("n"), resync_interval=3600)
+        ext.register_node("n", [16])
+        srv = ExtenderServer(ext).start()
+        try:
+            s = HttpSession(srv.url)
+            status, _ = s.request(
+                "POST", "/gpushare-scheduler/filter", body=b"{not json",
+                headers={"Content-Type": "application/json"},
+            )
+            assert status == 500
+            # extender still healthy
+            status, _ = s.request("GET", "/gpushare-scheduler/packing")
+            assert status == 200
+            s.close()
+        finally:
+            srv.stop()
