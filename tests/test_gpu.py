@@ -374,3 +374,56 @@ def test_memguard_budget_shared_across_processes():
     if os.path.exists(shm_path):
         os.unlink(shm_path)
     assert ok, "budget was not repaid after holder exit"
+
+
+@pytest.mark.gpu
+def test_eight_tenants_shared_gpu_with_memguard():
+    """The end goal of the framework: 8 co-located tenants on ONE MI355X,
+    each enforced to a 30 GiB budget, all running real GEMMs concurrently;
+    every tenant succeeds inside its budget and sees the clamped VRAM."""
+    import os
+    import subprocess
+    import sys
+    import uuid as uuid_mod
+
+    import gpushare_amd
+
+    lib = os.path.join(
+        os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+    )
+    script = (
+        "import os, torch;"
+        "free, total = torch.cuda.mem_get_info();"
+        "assert total <= 30<<30, f'not clamped: {total}';"
+        "x = torch.randn(8192, 8192, device='cuda:0', dtype=torch.bfloat16);"
+        "w = torch.randn(8192, 8192, device='cuda:0', dtype=torch.bfloat16);"
+        "y = (x @ w).float().sum(); torch.cuda.synchronize();"
+        "assert torch.isfinite(y);"
+        "print('TENANT_OK', os.environ['T_ID'], flush=True)"
+    )
+    procs = []
+    shm_paths = []
+    for i in range(8):
+        env = dict(os.environ)
+        pod_uid = f"tenant{i}-{uuid_mod.uuid4().hex[:8]}"
+        env["LD_PRELOAD"] = lib
+        env["GPUSHARE_MEM_LIMIT_BYTES"] = str(30 << 30)  # 8×30 < 288 GiB
+        env["GPUSHARE_POD_UID"] = pod_uid
+        env["T_ID"] = str(i)
+        shm_paths.append(f"/dev/shm/gpushare.memguard.{pod_uid}")
+        procs.append(
+            subprocess.Popen(
+                [sys.executable, "-c", script],
+                env=env, stdout=subprocess.PIPE,
+                stderr=subprocess.STDOUT, text=True,
+            )
+        )
+    failures = []
+    for i, p in enumerate(procs):
+        out, _ = p.communicate(timeout=300)
+        if p.returncode != 0 or "TENANT_OK" not in out:
+            failures.append((i, p.returncode, out[-1500:]))
+    for sp in shm_paths:
+        if os.path.exists(sp):
+            os.unlink(sp)
+    assert not failures, failures
