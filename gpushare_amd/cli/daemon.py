@@ -7,6 +7,9 @@ code; deliberately not carried), plus MI355X-native additions:
   --mock-spec        run against fake GPUs (CI; BASELINE config 1)
   --deep-probe       active gfx950 canary health probe interval (seconds)
   --no-inject        don't add /dev/kfd+renderD DeviceSpecs (env-only mode)
+  --no-informer      list-per-Allocate instead of the pod watch cache
+  --memguard-dir     stage + inject the LD_PRELOAD VRAM budget enforcer
+  --allow-oversize-inventory  serve >4MiB ListAndWatch (patched kubelets)
 
 Env: NODE_NAME (required — reference crashes at package-import time if
 unset, podmanager.go:52-55; we fail at startup with a clear message),
