@@ -121,3 +121,26 @@ def test_dedup_by_uid(kube):
 
     pm = _pm(kube, kubelet_client=DupKubelet())
     assert len(pm.get_pending_pods()) == 1
+
+
+def test_real_kubelet_client_over_http():
+    """KubeletClient (the real HTTP client, reference client.go:119-134)
+    against the fake apiserver's kubelet read-only /pods view."""
+    from gpushare_amd.cluster.fakeapiserver import FakeApiServer
+    from gpushare_amd.cluster.kubeclient import FakeKubeClient, KubeletClient
+
+    from helpers import make_pod
+
+    server = FakeApiServer(store=FakeKubeClient(node_name="node-a")).start()
+    try:
+        server.store.add_pod(make_pod("p1", node="node-a", mem=4))
+        server.store.add_pod(make_pod("elsewhere", node="node-b", mem=4))
+        kc = KubeletClient(
+            address="127.0.0.1", port=server.port, scheme="http", token="t"
+        )
+        pods = kc.get_node_running_pods()
+        names = [p["metadata"]["name"] for p in pods["items"]]
+        assert names == ["p1"]  # node-scoped view
+        kc.close()
+    finally:
+        server.stop()
