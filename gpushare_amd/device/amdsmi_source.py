@@ -124,6 +124,18 @@ class AmdSmiSource:
     def devices(self) -> list[PhysicalGPU]:
         return list(self._gpus)
 
+    def process_usage(self) -> dict[int, list]:
+        """Per-process VRAM/engine usage per GPU (amdsmi process list;
+        includes container_name where the driver resolves the cgroup) —
+        pod-level attribution for observability/debugging."""
+        out: dict[int, list] = {}
+        for g in self._gpus:
+            try:
+                out[g.index] = self._smi.process_list(g.index)
+            except RuntimeError:
+                out[g.index] = []
+        return out
+
     def vram_usage(self) -> dict[int, int]:
         """Live per-GPU VRAM used bytes (observability; the scheduling
         currency stays the *allocated* annotations, not live usage)."""

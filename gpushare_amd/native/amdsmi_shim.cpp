@@ -70,6 +70,8 @@ struct Api {
   amdsmi_status_t (*get_gpu_event_notification)(
       int, uint32_t *, amdsmi_evt_notification_data_t *);
   amdsmi_status_t (*stop_gpu_event_notification)(amdsmi_processor_handle);
+  amdsmi_status_t (*get_gpu_process_list)(amdsmi_processor_handle, uint32_t *,
+                                          amdsmi_proc_info_t *);
 } g_api;
 
 std::vector<amdsmi_processor_handle> g_procs;  // flattened across sockets
@@ -114,6 +116,7 @@ bool load_library() {
           "amdsmi_set_gpu_event_notification_mask");
   resolve(g_api.get_gpu_event_notification, "amdsmi_get_gpu_event_notification");
   resolve(g_api.stop_gpu_event_notification, "amdsmi_stop_gpu_event_notification");
+  resolve(g_api.get_gpu_process_list, "amdsmi_get_gpu_process_list");
   return true;
 }
 
@@ -176,6 +179,36 @@ void smi_shutdown() {
   g_api.shut_down();
   g_procs.clear();
   g_inited = false;
+}
+
+py::list process_list(size_t i) {
+  // Per-process VRAM/engine usage on one GPU (observability: pod-level
+  // attribution; NVML analogue nvmlDeviceGetComputeRunningProcesses has
+  // no call site in the reference).  amdsmi fills container_name from the
+  // process cgroup where available.
+  amdsmi_processor_handle h = proc(i);
+  if (g_api.get_gpu_process_list == nullptr)
+    throw std::runtime_error("amdsmi_get_gpu_process_list unavailable");
+  uint32_t n = 0;
+  amdsmi_status_t st = g_api.get_gpu_process_list(h, &n, nullptr);
+  py::list out;
+  if (st != AMDSMI_STATUS_SUCCESS || n == 0) return out;
+  std::vector<amdsmi_proc_info_t> procs(n);
+  std::memset(procs.data(), 0, sizeof(amdsmi_proc_info_t) * n);
+  st = g_api.get_gpu_process_list(h, &n, procs.data());
+  if (st != AMDSMI_STATUS_SUCCESS) return out;
+  for (uint32_t k = 0; k < n; ++k) {
+    py::dict p;
+    p["pid"] = (py::int_)procs[k].pid;
+    p["name"] = std::string(procs[k].name);
+    p["container_name"] = std::string(procs[k].container_name);
+    p["vram_bytes"] = (py::int_)procs[k].memory_usage.vram_mem;
+    p["gtt_bytes"] = (py::int_)procs[k].memory_usage.gtt_mem;
+    p["gfx_engine_ns"] = (py::int_)procs[k].engine_usage.gfx;
+    p["cu_occupancy"] = (py::int_)procs[k].cu_occupancy;
+    out.append(p);
+  }
+  return out;
 }
 
 size_t device_count() {
@@ -301,6 +334,7 @@ PYBIND11_MODULE(_amdsmi, m) {
   m.def("device_count", &device_count);
   m.def("device_info", &device_info, py::arg("index"));
   m.def("ecc_count", &ecc_count, py::arg("index"));
+  m.def("process_list", &process_list, py::arg("index"));
   m.def("event_watch_init", &event_watch_init, py::arg("index"), py::arg("mask"));
   m.def("event_poll", &event_poll, py::arg("timeout_ms") = 5000,
         py::arg("max_events") = 64);

@@ -427,3 +427,36 @@ def test_eight_tenants_shared_gpu_with_memguard():
         if os.path.exists(sp):
             os.unlink(sp)
     assert not failures, failures
+
+
+@pytest.mark.gpu
+def test_process_list_attribution(source):
+    """amdsmi process list: a live torch process on GPU 0 must appear with
+    nonzero VRAM (the raw data for pod-level usage attribution)."""
+    import subprocess
+    import sys
+    import time
+
+    script = (
+        "import torch, time;"
+        "x = torch.empty(1 << 28, dtype=torch.uint8, device='cuda:0');"
+        "print('UP', flush=True); time.sleep(30)"
+    )
+    proc = subprocess.Popen(
+        [sys.executable, "-c", script], stdout=subprocess.PIPE, text=True
+    )
+    try:
+        assert "UP" in proc.stdout.readline()
+        deadline = time.monotonic() + 20
+        found = None
+        while time.monotonic() < deadline and found is None:
+            for p in source.process_usage().get(0, []):
+                if p["pid"] == proc.pid:
+                    found = p
+                    break
+            time.sleep(1)
+        assert found is not None, "torch process not in amdsmi process list"
+        assert found["vram_bytes"] >= 1 << 28, found
+    finally:
+        proc.terminate()
+        proc.wait(timeout=30)
