@@ -127,7 +127,8 @@ class AmdSmiSource:
     def process_usage(self) -> dict[int, list]:
         """Per-process VRAM/engine usage per GPU (amdsmi process list;
         includes container_name where the driver resolves the cgroup) —
-        pod-level attribution for observability/debugging."""
+        pod-level attribution for observability/debugging.  Note: pids are
+        HOST-namespace (map to pods via /proc/<pid>/cgroup on the host)."""
         out: dict[int, list] = {}
         for g in self._gpus:
             try:

@@ -189,14 +189,20 @@ py::list process_list(size_t i) {
   amdsmi_processor_handle h = proc(i);
   if (g_api.get_gpu_process_list == nullptr)
     throw std::runtime_error("amdsmi_get_gpu_process_list unavailable");
-  uint32_t n = 0;
-  amdsmi_status_t st = g_api.get_gpu_process_list(h, &n, nullptr);
-  py::list out;
-  if (st != AMDSMI_STATUS_SUCCESS || n == 0) return out;
+  // single-call form: pass a generously sized buffer; amdsmi updates n
+  // to the number of processes written (the two-call count form is not
+  // reliable across amdsmi versions)
+  uint32_t n = 512;
   std::vector<amdsmi_proc_info_t> procs(n);
   std::memset(procs.data(), 0, sizeof(amdsmi_proc_info_t) * n);
-  st = g_api.get_gpu_process_list(h, &n, procs.data());
-  if (st != AMDSMI_STATUS_SUCCESS) return out;
+  amdsmi_status_t st = g_api.get_gpu_process_list(h, &n, procs.data());
+  py::list out;
+  if (st != AMDSMI_STATUS_SUCCESS) {
+    char buf[64];
+    snprintf(buf, sizeof(buf), "get_gpu_process_list failed: %d", (int)st);
+    throw std::runtime_error(buf);
+  }
+  if (n > procs.size()) n = procs.size();
   for (uint32_t k = 0; k < n; ++k) {
     py::dict p;
     p["pid"] = (py::int_)procs[k].pid;
