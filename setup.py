@@ -43,6 +43,7 @@ setup(
             "kubectl-inspect-gpushare=gpushare_amd.cli.inspect:main",
             "gpushare-podgetter=gpushare_amd.cli.podgetter:main",
             "gpushare-scheduler-extender=gpushare_amd.extender.__main__:main",
+            "gpushare-top=gpushare_amd.cli.top:main",
         ]
     },
 )

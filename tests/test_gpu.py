@@ -465,3 +465,17 @@ def test_process_list_attribution(source):
     finally:
         proc.terminate()
         proc.wait(timeout=30)
+
+
+@pytest.mark.gpu
+def test_gpushare_top_on_real_gpu(source):
+    """gpushare-top against the real device source: renders without error,
+    shows the MI355X VRAM total."""
+    import io
+
+    from gpushare_amd.cli import top
+
+    out = io.StringIO()
+    assert top.main([], source=source, out=out) == 0
+    text = out.getvalue()
+    assert "GiB" in text and "renderD" in text

@@ -114,3 +114,36 @@ def test_inspect_over_http_apiserver():
         assert "4/8" in out.getvalue()
     finally:
         api.stop()
+
+
+def test_gpushare_top_snapshot():
+    """gpushare-top renders per-GPU and per-process rows from a source
+    (mock source lacks usage APIs -> zero-usage snapshot still works)."""
+    import io
+
+    from gpushare_amd.cli import top
+    from gpushare_amd.device.mock_source import MockSource
+
+    class FakeUsageSource:
+        def __init__(self):
+            self._gpus = MockSource.from_spec("2x288GiB").devices()
+
+        def devices(self):
+            return self._gpus
+
+        def vram_usage(self):
+            return {0: 5 << 30, 1: 0}
+
+        def process_usage(self):
+            return {
+                0: [{"pid": 4242, "vram_bytes": 5 << 30, "gtt_bytes": 0,
+                     "cu_occupancy": 12, "container_name": "tenant-a",
+                     "gfx_engine_ns": 10}],
+                1: [],
+            }
+
+    out = io.StringIO()
+    assert top.main([], source=FakeUsageSource(), out=out) == 0
+    text = out.getvalue()
+    assert "5.0GiB/288.0GiB" in text
+    assert "4242" in text and "tenant-a" in text
