@@ -247,3 +247,65 @@ class TestDegradedMode:
         resp = alloc.allocate(_request([8]))
         envs = resp.container_responses[0].envs
         assert envs[consts.ENV_RESOURCE_INDEX] == "1"
+
+
+class TestInformerModelCheck:
+    def test_random_op_sequences_converge_to_server_state(self, api):
+        """Model-based check: after any seeded sequence of creates,
+        patches and deletes, the informer store converges to exactly the
+        server's node-scoped pod set."""
+        import random
+
+        inf = make_informer(api).start()
+        assert inf.wait_synced(5)
+        try:
+            rng = random.Random(1234)
+            live: set[str] = set()
+            counter = 0
+            for _ in range(300):
+                op = rng.random()
+                if op < 0.45 or not live:
+                    name = f"p{counter}"
+                    counter += 1
+                    node = NODE if rng.random() < 0.8 else "node-b"
+                    api.store.add_pod(make_pod(name, node=node, mem=4))
+                    if node == NODE:
+                        live.add(name)
+                elif op < 0.75:
+                    name = rng.choice(sorted(live))
+                    api.store.patch_pod(
+                        "default", name,
+                        {"metadata": {"annotations": {"seq": str(counter)}}},
+                    )
+                    counter += 1
+                else:
+                    name = rng.choice(sorted(live))
+                    api.store.delete_pod("default", name)
+                    live.discard(name)
+            assert wait_for(
+                lambda: {
+                    p["metadata"]["name"] for p in inf.pods()
+                } == live,
+                timeout=10,
+            ), (
+                f"store diverged: informer="
+                f"{sorted(p['metadata']['name'] for p in inf.pods())[:10]}... "
+                f"expected {len(live)} pods"
+            )
+            # annotations converge too (latest resourceVersion wins)
+            server_rv = {
+                name: api.store.pods[("default", name)]["metadata"][
+                    "resourceVersion"
+                ]
+                for name in live
+            }
+            inf_rv = {
+                p["metadata"]["name"]: p["metadata"]["resourceVersion"]
+                for p in inf.pods()
+            }
+            assert wait_for(lambda: {
+                p["metadata"]["name"]: p["metadata"]["resourceVersion"]
+                for p in inf.pods()
+            } == server_rv), f"rv mismatch: {inf_rv} != {server_rv}"
+        finally:
+            inf.stop()
