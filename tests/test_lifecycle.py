@@ -175,3 +175,50 @@ def test_daemon_cli_flag_validation():
     assert args.query_kubelet
     with pytest.raises(SystemExit):
         parse_args(["--memory-unit", "TiB"])
+
+
+def test_allocation_state_survives_plugin_restart(tmp_socket_dir):
+    """Crash-only invariant (reference §3.5): a pod assumed BEFORE a plugin
+    restart allocates correctly AFTER it — all state lives in annotations,
+    and a pod assigned before the restart is not re-matched after."""
+    from helpers import make_pod
+
+    kube, source, mgr = _manager(tmp_socket_dir, spec="2x16GiB")
+    kubelet = StubKubelet(tmp_socket_dir)
+    kubelet.start()
+    t = threading.Thread(target=mgr.run, daemon=True)
+    t.start()
+    try:
+        client = kubelet.wait_for_plugin(consts.RESOURCE_NAME)
+        ids = sorted(client.wait_for_devices(32))
+
+        # pod A allocated pre-restart
+        kube.add_pod(make_pod("pre-a", 4, gpu_idx=0, node="node-a"))
+        resp = client.allocate([ids[:4]])
+        assert resp.container_responses[0].envs[consts.ENV_RESOURCE_INDEX] == "0"
+        # pod B assumed pre-restart, allocated post-restart
+        kube.add_pod(make_pod("pre-b", 6, gpu_idx=1, node="node-a"))
+
+        mgr._signals.put(signal.SIGHUP)
+        first_plugin = mgr.plugin
+        deadline = time.monotonic() + 10
+        while mgr.plugin is first_plugin and time.monotonic() < deadline:
+            time.sleep(0.05)
+        client = kubelet.wait_for_plugin(consts.RESOURCE_NAME, timeout=10)
+
+        resp = client.allocate([ids[:6]])
+        envs = resp.container_responses[0].envs
+        assert envs[consts.ENV_RESOURCE_INDEX] == "1"  # pre-b, not pre-a
+        anns = kube.get_pod("default", "pre-b")["metadata"]["annotations"]
+        assert anns[consts.ENV_ASSIGNED_FLAG] == "true"
+        # pre-a stays assigned exactly once (not re-matched)
+        assert (
+            kube.get_pod("default", "pre-a")["metadata"]["annotations"][
+                consts.ENV_ASSIGNED_FLAG
+            ]
+            == "true"
+        )
+    finally:
+        mgr.shutdown()
+        t.join(timeout=5)
+        kubelet.stop()
