@@ -41,7 +41,7 @@ class FakeDeviceTable:
     gpu_of: dict[str, int]         # fake id -> plugin GPU index
     uuid_of: dict[int, str]        # GPU index -> uuid
     index_of: dict[str, int]       # uuid -> GPU index
-    ranges: list[tuple[int, int]]  # per GPU: [start, end) into ids
+    ranges: dict[int, tuple[int, int]]  # GPU index -> [start, end) into ids
     unit: str
     units_per_gpu: list[int]
 
@@ -51,7 +51,7 @@ class FakeDeviceTable:
             raise ValueError(f"invalid memory unit {unit!r}")
         ids: list[str] = []
         gpu_of: dict[str, int] = {}
-        ranges: list[tuple[int, int]] = []
+        ranges: dict[int, tuple[int, int]] = {}
         units_per_gpu: list[int] = []
         for g in gpus:
             n = g.mem_units(unit)
@@ -62,7 +62,9 @@ class FakeDeviceTable:
                 fid = fake_id(g.uuid, j)
                 ids.append(fid)
                 gpu_of[fid] = g.index
-            ranges.append((start, len(ids)))
+            # keyed by the PLUGIN index, not list position — robust to
+            # non-contiguous index sets (e.g. partitioned/filtered nodes)
+            ranges[g.index] = (start, len(ids))
             units_per_gpu.append(n)
         return cls(
             ids=ids,
