@@ -1,0 +1,100 @@
+#!/usr/bin/env python3
+"""Co-location fairness on one MI355X: what sharing actually costs.
+
+gpushare's premise is N pods space/time-sharing one GPU.  This experiment
+measures per-tenant GEMM throughput as co-location density grows — the
+number a capacity planner needs when deciding pods-per-GPU.
+
+For N in {1, 2, 4, 8}: launch N identical tenant processes on GPU 0
+(each under a memguard budget as Allocate would inject), each running
+bf16 GEMMs for a fixed window; report per-tenant and aggregate TFLOP/s
+plus the fairness spread (min/max across tenants).
+
+Usage (on an MI355X node):  python benchmarks/colocation_fairness.py
+Prints one JSON line per density.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+TENANT = r"""
+import json, os, time
+import torch
+n = 8192
+x = torch.randn(n, n, device="cuda:0", dtype=torch.bfloat16)
+w = torch.randn(n, n, device="cuda:0", dtype=torch.bfloat16)
+for _ in range(3):
+    y = x @ w
+torch.cuda.synchronize()
+WINDOW = float(os.environ.get("T_WINDOW", "8"))
+t0 = time.perf_counter()
+iters = 0
+while time.perf_counter() - t0 < WINDOW:
+    y = x @ w
+    iters += 1
+    if iters % 8 == 0:
+        torch.cuda.synchronize()
+torch.cuda.synchronize()
+dt = time.perf_counter() - t0
+tflops = iters * (2 * n**3) / dt / 1e12
+print(json.dumps({"iters": iters, "secs": round(dt, 3),
+                  "tflops": round(tflops, 1)}), flush=True)
+"""
+
+
+def run_density(n_tenants: int, budget_gib: int = 30) -> dict:
+    import gpushare_amd
+
+    lib = os.path.join(
+        os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+    )
+    procs = []
+    for i in range(n_tenants):
+        env = dict(os.environ)
+        if os.path.exists(lib):
+            env["LD_PRELOAD"] = lib
+            env["GPUSHARE_MEM_LIMIT_BYTES"] = str(budget_gib << 30)
+            env["GPUSHARE_POD_UID"] = f"fair{n_tenants}-{i}"
+        procs.append(
+            subprocess.Popen(
+                [sys.executable, "-c", TENANT],
+                env=env, stdout=subprocess.PIPE,
+                stderr=subprocess.PIPE, text=True, cwd=REPO,
+            )
+        )
+    results = []
+    for p in procs:
+        out, err = p.communicate(timeout=300)
+        if p.returncode != 0:
+            raise RuntimeError(f"tenant failed: {err[-800:]}")
+        results.append(json.loads(out.strip().splitlines()[-1]))
+    for i in range(n_tenants):
+        shm = f"/dev/shm/gpushare.memguard.fair{n_tenants}-{i}"
+        if os.path.exists(shm):
+            os.unlink(shm)
+    tfl = [r["tflops"] for r in results]
+    return {
+        "tenants": n_tenants,
+        "per_tenant_tflops": tfl,
+        "aggregate_tflops": round(sum(tfl), 1),
+        "min_tflops": min(tfl),
+        "max_tflops": max(tfl),
+        "fairness_min_over_max": round(min(tfl) / max(tfl), 3),
+    }
+
+
+def main() -> int:
+    sys.path.insert(0, REPO)
+    for n in (1, 2, 4, 8):
+        print(json.dumps(run_density(n)), flush=True)
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
