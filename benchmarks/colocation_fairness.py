@@ -89,10 +89,67 @@ def run_density(n_tenants: int, budget_gib: int = 30) -> dict:
     }
 
 
+BW_TENANT = r"""
+import json, os, time
+import torch
+n_bytes = 4 << 30
+src = torch.empty(n_bytes, dtype=torch.uint8, device="cuda:0")
+dst = torch.empty(n_bytes, dtype=torch.uint8, device="cuda:0")
+dst.copy_(src); torch.cuda.synchronize()
+WINDOW = float(os.environ.get("T_WINDOW", "8"))
+t0 = time.perf_counter()
+iters = 0
+while time.perf_counter() - t0 < WINDOW:
+    dst.copy_(src)
+    iters += 1
+    if iters % 4 == 0:
+        torch.cuda.synchronize()
+torch.cuda.synchronize()
+dt = time.perf_counter() - t0
+gbps = iters * 2 * n_bytes / dt / 1e9   # read + write traffic
+print(json.dumps({"iters": iters, "secs": round(dt, 3),
+                  "hbm_gbps": round(gbps, 1)}), flush=True)
+"""
+
+
+def run_mixed() -> dict:
+    """Cross-workload interference: a compute-bound GEMM tenant and an
+    HBM-bandwidth-bound copy tenant, solo then co-located."""
+    def run(scripts):
+        procs = [
+            subprocess.Popen([sys.executable, "-c", s],
+                             stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+                             text=True, cwd=REPO)
+            for s in scripts
+        ]
+        out = []
+        for p in procs:
+            o, e = p.communicate(timeout=300)
+            if p.returncode != 0:
+                raise RuntimeError(e[-800:])
+            out.append(json.loads(o.strip().splitlines()[-1]))
+        return out
+
+    gemm_solo = run([TENANT])[0]
+    bw_solo = run([BW_TENANT])[0]
+    gemm_co, bw_co = run([TENANT, BW_TENANT])
+    return {
+        "mixed_interference": {
+            "gemm_solo_tflops": gemm_solo["tflops"],
+            "gemm_colocated_tflops": gemm_co["tflops"],
+            "gemm_retained": round(gemm_co["tflops"] / gemm_solo["tflops"], 3),
+            "bw_solo_gbps": bw_solo["hbm_gbps"],
+            "bw_colocated_gbps": bw_co["hbm_gbps"],
+            "bw_retained": round(bw_co["hbm_gbps"] / bw_solo["hbm_gbps"], 3),
+        }
+    }
+
+
 def main() -> int:
     sys.path.insert(0, REPO)
     for n in (1, 2, 4, 8):
         print(json.dumps(run_density(n)), flush=True)
+    print(json.dumps(run_mixed()), flush=True)
     return 0
 
 
