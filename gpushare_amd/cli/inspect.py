@@ -64,6 +64,21 @@ class NodeInfo:
             if addr.get("type") == "InternalIP":
                 self.address = addr.get("address", "unknown")
                 break
+        # real per-GPU capacities from the plugin's topology annotation
+        # (heterogeneous nodes; reference assumed total/count everywhere)
+        self.per_gpu_units: list[int] = []
+        raw = (node.get("metadata", {}).get("annotations") or {}).get(
+            consts.ANN_NODE_TOPOLOGY
+        )
+        if raw:
+            import json as _json
+
+            try:
+                units = _json.loads(raw).get("per_gpu_units", [])
+                if len(units) == self.gpu_count:
+                    self.per_gpu_units = [int(u) for u in units]
+            except (ValueError, TypeError):
+                pass
         # devs: idx -> {"used": int, "pods": [pod]}
         self.devs: dict[int, dict] = {
             i: {"used": 0, "pods": []} for i in range(self.gpu_count)
@@ -79,6 +94,11 @@ class NodeInfo:
     @property
     def per_gpu_total(self) -> int:
         return self.total_mem // self.gpu_count if self.gpu_count else 0
+
+    def gpu_total(self, idx: int) -> int:
+        if 0 <= idx < len(self.per_gpu_units):
+            return self.per_gpu_units[idx]
+        return self.per_gpu_total
 
     @property
     def used_mem(self) -> int:
@@ -129,7 +149,7 @@ def display_summary(infos: list[NodeInfo], out=sys.stdout) -> None:
         for i in range(max_gpus):
             dev = n.devs.get(i)
             row.append(
-                f"{dev['used']}/{n.per_gpu_total}" if dev is not None and i < n.gpu_count
+                f"{dev['used']}/{n.gpu_total(i)}" if dev is not None and i < n.gpu_count
                 else "0/0"
             )
         if has_pending:

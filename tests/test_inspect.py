@@ -147,3 +147,39 @@ def test_gpushare_top_snapshot():
     text = out.getvalue()
     assert "5.0GiB/288.0GiB" in text
     assert "4242" in text and "tenant-a" in text
+
+
+def test_summary_heterogeneous_capacities():
+    """With the topology annotation present, the summary shows each GPU's
+    REAL capacity instead of total/count (reference behavior)."""
+    import io
+    import json as _json
+
+    from gpushare_amd.cli import inspect as insp
+
+    node = {
+        "metadata": {
+            "name": "hetero",
+            "annotations": {
+                consts.ANN_NODE_TOPOLOGY: _json.dumps(
+                    {"unit": "GiB", "per_gpu_units": [288, 96],
+                     "xgmi": [[1], [0]]}
+                )
+            },
+        },
+        "status": {
+            "allocatable": {
+                consts.RESOURCE_COUNT: "2",
+                consts.RESOURCE_NAME: "384",
+            },
+            "addresses": [{"type": "InternalIP", "address": "10.0.0.9"}],
+        },
+    }
+    pod = make_pod("p", mem=64, gpu_idx=0, node="hetero")
+    pod["metadata"]["annotations"][consts.ENV_ASSIGNED_FLAG] = "true"
+    info = insp.NodeInfo(node, [pod])
+    out = io.StringIO()
+    insp.display_summary([info], out=out)
+    text = out.getvalue()
+    assert "64/288" in text
+    assert "0/96" in text
