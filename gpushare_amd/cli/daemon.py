@@ -130,6 +130,8 @@ def main(argv=None) -> int:
             port=args.kubelet_port,
             token=args.token or None,
             timeout=args.timeout,
+            client_cert=args.client_cert,
+            client_key=args.client_key,
         )
 
     mgr = SharedGPUManager(

@@ -246,8 +246,10 @@ class KubeletClient:
         token: Optional[str] = None,
         timeout: float = 10.0,
         scheme: str = "https",
+        client_cert: str = "",
+        client_key: str = "",
     ):
-        from .httpconn import HttpSession
+        from .httpconn import HttpSession, make_ssl_context
 
         headers = {"Accept": "application/json"}
         if token is None:
@@ -256,10 +258,18 @@ class KubeletClient:
                 token = open(token_path).read().strip()
         if token:
             headers["Authorization"] = f"Bearer {token}"
+        # mTLS option (reference: --client-cert/--client-key flags feed the
+        # kubelet client, cmd/nvidia/main.go:28-53); server verification
+        # stays off either way (client.go:75-99 forces insecure)
+        verify = False
+        if client_cert:
+            ctx = make_ssl_context(False)
+            ctx.load_cert_chain(client_cert, client_key or None)
+            verify = ctx
         self._client = HttpSession(
             f"{scheme}://{address}:{port}",
             headers=headers,
-            verify=False,
+            verify=verify,
             timeout=timeout,
         )
 

@@ -41,6 +41,7 @@ setup(
         "console_scripts": [
             "amdgpushare-device-plugin=gpushare_amd.cli.daemon:main",
             "kubectl-inspect-gpushare=gpushare_amd.cli.inspect:main",
+            "kubectl-inspect-gpushare-v2=gpushare_amd.cli.inspect:main",
             "gpushare-podgetter=gpushare_amd.cli.podgetter:main",
             "gpushare-scheduler-extender=gpushare_amd.extender.__main__:main",
             "gpushare-top=gpushare_amd.cli.top:main",
