@@ -14,6 +14,9 @@ RUN python3 -m pip install --no-cache-dir pybind11 && \
 FROM rocm/rocm-terminal:7.2
 RUN python3 -m pip install --no-cache-dir grpcio protobuf pyyaml prometheus_client
 COPY --from=build /usr/local/lib/python3*/dist-packages /usr/local/lib/python3/dist-packages
-COPY --from=build /usr/local/bin/amdgpushare-device-plugin /usr/local/bin/
+COPY --from=build /usr/local/bin/amdgpushare-device-plugin \
+                  /usr/local/bin/gpushare-scheduler-extender \
+                  /usr/local/bin/gpushare-top \
+                  /usr/local/bin/kubectl-inspect-gpushare /usr/local/bin/
 ENV HSA_ENABLE_IPC_MODE_LEGACY=0
 ENTRYPOINT ["amdgpushare-device-plugin"]
