@@ -91,3 +91,33 @@ def test_concurrent_assume_never_overcommits(caps):
     for i, cap in enumerate(caps):
         assert st_node.allocated[i] <= cap
     assert sum(sum(s.values()) for s in results) == st_node.total_allocated
+
+
+# ----------------------------------------------------------------------- #
+# devlist codec: native encoder must be byte-identical to the python oracle
+# ----------------------------------------------------------------------- #
+
+ids_strategy = st.lists(
+    st.text(
+        alphabet="abcdef0123456789-", min_size=1, max_size=40
+    ).map(lambda s: s + "-_-0"),
+    min_size=0,
+    max_size=50,
+)
+
+
+@given(ids=ids_strategy, data=st.data())
+@settings(max_examples=100, deadline=None)
+def test_native_codec_matches_python_oracle(ids, data):
+    from gpushare_amd.device.fakedev import encode_list_python, make_codec
+
+    unhealthy = set(
+        data.draw(
+            st.lists(
+                st.integers(min_value=0, max_value=max(len(ids) - 1, 0)),
+                max_size=len(ids),
+            )
+        )
+    ) if ids else set()
+    codec = make_codec(ids)
+    assert codec.encode(sorted(unhealthy)) == encode_list_python(ids, unhealthy)
