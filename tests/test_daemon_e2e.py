@@ -1,0 +1,97 @@
+"""Daemon entrypoint e2e: the REAL production startup path.
+
+`python -m gpushare_amd.cli.daemon` as a subprocess with KUBECONFIG
+pointing at the fake apiserver and a stub kubelet on the socket dir —
+covers kubeconfig parsing (RestKubeClient._auto_config), mock-source
+startup, node patching (gpu-count + topology annotation), registration,
+ListAndWatch, and graceful SIGTERM shutdown.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import signal
+import subprocess
+import sys
+import time
+
+import pytest
+
+from gpushare_amd import consts
+from gpushare_amd.cluster.fakeapiserver import FakeApiServer
+from gpushare_amd.cluster.kubeclient import FakeKubeClient
+from gpushare_amd.deviceplugin.stubkubelet import StubKubelet
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+NODE = "daemon-e2e-node"
+
+
+@pytest.fixture()
+def api():
+    store = FakeKubeClient(node_name=NODE)
+    store.nodes = {
+        NODE: {
+            "metadata": {"name": NODE, "labels": {}},
+            "status": {"capacity": {}, "allocatable": {}},
+        }
+    }
+    server = FakeApiServer(store=store).start()
+    yield server
+    server.stop()
+
+
+def test_daemon_main_with_kubeconfig(api, tmp_path):
+    kubeconfig = tmp_path / "kubeconfig"
+    kubeconfig.write_text(
+        json.dumps(  # valid YAML (JSON subset)
+            {
+                "current-context": "e2e",
+                "contexts": [
+                    {"name": "e2e",
+                     "context": {"cluster": "c", "user": "u"}}
+                ],
+                "clusters": [
+                    {"name": "c", "cluster": {"server": api.url}}
+                ],
+                "users": [{"name": "u", "user": {"token": "test-token"}}],
+            }
+        )
+    )
+    sockdir = tmp_path / "dp"
+    sockdir.mkdir()
+    kubelet = StubKubelet(str(sockdir))
+    kubelet.start()
+    env = dict(os.environ)
+    env.update(
+        NODE_NAME=NODE,
+        KUBECONFIG=str(kubeconfig),
+        PYTHONPATH=REPO,
+    )
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "gpushare_amd.cli.daemon",
+         "--mock-spec", "2x16GiB", "--socket-dir", str(sockdir)],
+        env=env, cwd=REPO,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+    )
+    try:
+        client = kubelet.wait_for_plugin(consts.RESOURCE_NAME, timeout=30)
+        devices = client.wait_for_devices(min_count=32, timeout=15)
+        assert len(devices) == 32
+
+        node = api.store.get_node(NODE)
+        assert node["status"]["capacity"][consts.RESOURCE_COUNT] == "2"
+        topo = json.loads(
+            node["metadata"]["annotations"][consts.ANN_NODE_TOPOLOGY]
+        )
+        assert topo["per_gpu_units"] == [16, 16]
+
+        proc.send_signal(signal.SIGTERM)
+        assert proc.wait(timeout=15) == 0, "daemon did not exit cleanly"
+    finally:
+        if proc.poll() is None:
+            proc.kill()
+            proc.wait(timeout=10)
+        kubelet.stop()
+    # socket unlinked on graceful stop
+    assert not os.path.exists(sockdir / consts.SERVER_SOCK_NAME)
