@@ -41,7 +41,9 @@ class FastHTTPServer:
         handler: Callable,
         port: int = 0,
         host: str = "127.0.0.1",
+        ssl_context=None,   # server-side TLS (kubelet-style HTTPS endpoints)
     ):
+        self._ssl_context = ssl_context
         import errno
         import time
 
@@ -99,6 +101,8 @@ class FastHTTPServer:
             except OSError:
                 return
             conn.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+            # (TLS handshake, if configured, happens in the per-connection
+            # thread — it may block)
             with self._conn_lock:
                 self._conns = {c for c in self._conns if c.fileno() != -1}
                 self._conns.add(conn)
@@ -107,6 +111,15 @@ class FastHTTPServer:
             ).start()
 
     def _serve_conn(self, conn: socket.socket) -> None:
+        if self._ssl_context is not None:
+            try:
+                conn = self._ssl_context.wrap_socket(conn, server_side=True)
+            except (OSError, ValueError):
+                try:
+                    conn.close()
+                except OSError:
+                    pass
+                return
         f = conn.makefile("rb", buffering=65536)
         try:
             while not self._stop.is_set():

@@ -176,3 +176,51 @@ class TestFastHTTPRobustness:
             s.close()
         finally:
             srv.stop()
+
+
+class TestTLS:
+    def test_https_session_and_kubelet_client(self, tmp_path):
+        """Production kubelet mode: HTTPS with a self-signed serving cert
+        and verification off (reference forces insecure, client.go:75-99).
+        Covers make_ssl_context + the HTTPS branch of HttpSession and
+        KubeletClient's default scheme."""
+        import json
+        import ssl
+        import subprocess
+
+        from gpushare_amd.cluster.fasthttp import FastHTTPServer
+        from gpushare_amd.cluster.httpconn import HttpSession
+        from gpushare_amd.cluster.kubeclient import KubeletClient
+
+        cert = tmp_path / "tls.crt"
+        key = tmp_path / "tls.key"
+        subprocess.run(
+            ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+             "-days", "1", "-subj", "/CN=127.0.0.1",
+             "-keyout", str(key), "-out", str(cert)],
+            check=True, capture_output=True,
+        )
+        ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+        ctx.load_cert_chain(str(cert), str(key))
+
+        podlist = {"kind": "PodList", "items": [
+            {"metadata": {"name": "p1", "namespace": "default"},
+             "spec": {"nodeName": "n"}, "status": {"phase": "Running"}}
+        ]}
+
+        def handler(method, path, body):
+            assert path.startswith("/pods")
+            return 200, json.dumps(podlist).encode()
+
+        srv = FastHTTPServer(handler, ssl_context=ctx).start()
+        try:
+            s = HttpSession(f"https://127.0.0.1:{srv.port}", verify=False)
+            status, body = s.request("GET", "/pods/")
+            assert status == 200 and json.loads(body) == podlist
+            s.close()
+
+            kc = KubeletClient(address="127.0.0.1", port=srv.port, token="t")
+            assert kc.get_node_running_pods() == podlist
+            kc.close()
+        finally:
+            srv.stop()
