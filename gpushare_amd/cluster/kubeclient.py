@@ -1,11 +1,14 @@
 """Minimal Kubernetes REST client + in-memory fake.
 
-The plugin needs four apiserver verbs (reference uses client-go for the
-same surface, pkg/gpu/nvidia/podmanager.go):
+Apiserver surface (reference uses client-go for the same core verbs,
+pkg/gpu/nvidia/podmanager.go):
   - GET  node                              (isolation label check)
+  - PATCH node                             (topology annotation)
   - PATCH node /status                     (aliyun.com/gpu-count)
   - LIST pods (fieldSelector, all ns)      (pending-pod fallback path)
+  - WATCH pods (chunked stream)            (informer fast path)
   - PATCH pod (strategic merge)            (ASSIGNED=true handshake)
+  - POST/DELETE pod, POST event            (bench harness / observability)
 
 plus the kubelet read-only endpoint:
   - GET https://<node>:10250/pods/         (primary pending-pod path,
