@@ -224,3 +224,28 @@ class TestTLS:
             kc.close()
         finally:
             srv.stop()
+
+
+def test_deploy_manifests_parse_and_reference_contract():
+    """Every deploy manifest parses as YAML and carries the wire-contract
+    names (resource, socket dir, NODE_NAME downward API)."""
+    import glob
+    import os
+
+    import yaml
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    files = sorted(glob.glob(os.path.join(repo, "deploy", "**", "*.yaml"),
+                             recursive=True))
+    assert len(files) >= 6
+    docs = {}
+    for f in files:
+        docs[f] = [d for d in yaml.safe_load_all(open(f)) if d]
+        assert docs[f], f
+
+    blob = "\n".join(open(f).read() for f in files)
+    assert consts.RESOURCE_NAME in blob
+    assert "/var/lib/kubelet/device-plugins" in blob
+    assert "NODE_NAME" in blob
+    assert "prioritizeVerb" in blob       # scheduler policy wiring
+    assert "/var/lib/gpushare" in blob    # memguard hostPath
