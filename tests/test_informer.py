@@ -309,3 +309,31 @@ class TestInformerModelCheck:
             } == server_rv), f"rv mismatch: {inf_rv} != {server_rv}"
         finally:
             inf.stop()
+
+
+class TestWatchCycling:
+    def test_informer_survives_rapid_stream_cycling(self, api):
+        """Real apiservers close watches periodically (timeoutSeconds);
+        the informer must stay convergent when the stream is cycled many
+        times while pods churn."""
+        inf = make_informer(api, reconnect_backoff=0.01).start()
+        assert inf.wait_synced(5)
+        try:
+            live = set()
+            for i in range(10):
+                name = f"cyc{i}"
+                api.store.add_pod(make_pod(name, node=NODE, mem=4))
+                live.add(name)
+                if i % 2 == 0:
+                    api.store.watch_close_all()   # server-side watch close
+                if i % 3 == 2:
+                    victim = sorted(live)[0]
+                    api.store.delete_pod("default", victim)
+                    live.discard(victim)
+            assert wait_for(
+                lambda: {p["metadata"]["name"] for p in inf.pods()} == live,
+                timeout=15,
+            )
+            assert inf.reconnects >= 1
+        finally:
+            inf.stop()
