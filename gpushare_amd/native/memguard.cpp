@@ -285,6 +285,26 @@ int hipMallocAsync(void** ptr, size_t size, void* stream) {
     return rc;
 }
 
+int hipMallocPitch(void** ptr, size_t* pitch, size_t width, size_t height) {
+    using PitchFn = int (*)(void**, size_t*, size_t, size_t);
+    static PitchFn fn = real<PitchFn>("hipMallocPitch");
+    if (fn == nullptr) return HIP_ERROR_INVALID_VALUE;
+    // the true footprint (pitch × height) is only known after the call:
+    // allocate first, account after, undo if over budget
+    int rc = fn(ptr, pitch, width, height);
+    if (rc != HIP_SUCCESS) return rc;
+    init_limit();
+    if (g_limit < 0) return rc;
+    size_t size = (*pitch) * height;
+    if (!reserve(size)) {
+        static FreeFn freer = real<FreeFn>("hipFree");
+        if (freer != nullptr) freer(*ptr);
+        return HIP_ERROR_OOM;
+    }
+    track(*ptr, size);
+    return rc;
+}
+
 int hipFree(void* ptr) {
     static FreeFn fn = real<FreeFn>("hipFree");
     if (fn == nullptr) return HIP_ERROR_INVALID_VALUE;
