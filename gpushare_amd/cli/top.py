@@ -74,6 +74,8 @@ def main(argv=None, source=None, out=sys.stdout) -> int:
     )
     p.add_argument("-w", "--watch", type=float, default=0.0, metavar="SECONDS",
                    help="refresh every N seconds (0 = one snapshot)")
+    p.add_argument("-o", "--output", choices=("table", "json"),
+                   default="table", help="output format")
     args = p.parse_args(argv)
 
     if source is None:
@@ -86,11 +88,38 @@ def main(argv=None, source=None, out=sys.stdout) -> int:
             return 1
     try:
         while True:
-            snapshot(source, out)
+            if args.output == "json":
+                import json
+
+                print(
+                    json.dumps(
+                        {
+                            "gpus": [
+                                {
+                                    "index": g.index,
+                                    "uuid": g.uuid,
+                                    "vram_total_bytes": g.memory_bytes,
+                                    "vram_used_bytes": source.vram_usage().get(
+                                        g.index, 0
+                                    ),
+                                    "render_path": g.render_path,
+                                    "processes": source.process_usage().get(
+                                        g.index, []
+                                    ),
+                                }
+                                for g in source.devices()
+                            ]
+                        }
+                    ),
+                    file=out,
+                )
+            else:
+                snapshot(source, out)
             if args.watch <= 0:
                 return 0
             time.sleep(args.watch)
-            print("\x1b[2J\x1b[H", end="", file=out)  # clear screen
+            if args.output == "table":
+                print("\x1b[2J\x1b[H", end="", file=out)  # clear screen
     except KeyboardInterrupt:
         return 0
 

@@ -183,3 +183,29 @@ def test_summary_heterogeneous_capacities():
     text = out.getvalue()
     assert "64/288" in text
     assert "0/96" in text
+
+
+def test_gpushare_top_json_output():
+    import io
+    import json
+
+    from gpushare_amd.cli import top
+    from gpushare_amd.device.mock_source import MockSource
+
+    class Src:
+        def __init__(self):
+            self._g = MockSource.from_spec("1x16GiB").devices()
+
+        def devices(self):
+            return self._g
+
+        def vram_usage(self):
+            return {0: 1 << 30}
+
+        def process_usage(self):
+            return {0: []}
+
+    out = io.StringIO()
+    assert top.main(["-o", "json"], source=Src(), out=out) == 0
+    d = json.loads(out.getvalue())
+    assert d["gpus"][0]["vram_used_bytes"] == 1 << 30
