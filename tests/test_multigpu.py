@@ -354,3 +354,37 @@ class TestHeterogeneousNode:
         del pod["metadata"]["annotations"]
         kube.add_pod(pod)
         assert ext.assume(pod, NODE) is None  # 96 < 100 and others full
+
+
+class TestTopologyAnnotationFallback:
+    def test_corrupt_annotation_falls_back_to_uniform(self):
+        """A broken topology annotation must not break discovery — fall
+        back to the reference's uniform total/count split."""
+        kube = FakeKubeClient(node_name=NODE)
+        kube.patch_node(NODE, {"metadata": {"annotations": {
+            consts.ANN_NODE_TOPOLOGY: "{not json",
+        }}})
+        kube.patch_node_status(NODE, {"status": {"allocatable": {
+            consts.RESOURCE_COUNT: "2", consts.RESOURCE_NAME: "32",
+        }}})
+        from gpushare_amd.extender.__main__ import discover_nodes
+
+        ext = GPUShareExtender(kube, resync_interval=3600)
+        assert discover_nodes(kube, ext) == 1
+        assert ext.state.nodes[NODE].per_gpu_units == [16, 16]
+
+    def test_wrong_length_annotation_ignored(self):
+        kube = FakeKubeClient(node_name=NODE)
+        kube.patch_node(NODE, {"metadata": {"annotations": {
+            consts.ANN_NODE_TOPOLOGY: json.dumps(
+                {"per_gpu_units": [16, 16, 16], "xgmi": []}
+            ),
+        }}})
+        kube.patch_node_status(NODE, {"status": {"allocatable": {
+            consts.RESOURCE_COUNT: "2", consts.RESOURCE_NAME: "32",
+        }}})
+        from gpushare_amd.extender.__main__ import discover_nodes
+
+        ext = GPUShareExtender(kube, resync_interval=3600)
+        assert discover_nodes(kube, ext) == 1
+        assert ext.state.nodes[NODE].per_gpu_units == [16, 16]
