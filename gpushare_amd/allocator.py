@@ -92,6 +92,7 @@ class Allocator:
         inject_devices: bool = True,
         event_recorder=None,
         memguard_path: str = "",
+        trace_file: str = "",
     ):
         self.gpus = {g.index: g for g in gpus}
         self.pods = pod_manager
@@ -104,6 +105,10 @@ class Allocator:
         # — the MI355X answer to the reference's closed-source cGPU module
         self.memguard_path = memguard_path
         self.stats = AllocateStats()
+        # optional JSONL trace of every Allocate (the
+        # reference has no tracing although Allocate latency is the
+        # north-star metric); one line per call, append-only
+        self._trace = open(trace_file, "a") if trace_file else None
         # Matching runs under a short in-memory critical section; the
         # ASSIGNED patch happens OUTSIDE it.  A matched pod is "claimed"
         # (uid -> expiry) so concurrent Allocates skip it; claims are
@@ -166,7 +171,20 @@ class Allocator:
             )
             return self._err_response(request, req_units)
         finally:
-            self.stats.record(time.perf_counter() - t0, list_t, patch_t, ok)
+            total = time.perf_counter() - t0
+            self.stats.record(total, list_t, patch_t, ok)
+            if self._trace is not None:
+                import json as _json
+
+                self._trace.write(_json.dumps({
+                    "ts": time.time(),
+                    "req_units": req_units,
+                    "ok": ok,
+                    "total_ms": round(total * 1e3, 3),
+                    "match_ms": round(list_t * 1e3, 3),
+                    "patch_ms": round(patch_t * 1e3, 3),
+                }) + "\n")
+                self._trace.flush()
 
     # ------------------------------------------------------------------ #
     def _match_and_claim(self, req_units: int) -> Optional[dict]:

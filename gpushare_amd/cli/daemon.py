@@ -74,6 +74,9 @@ def parse_args(argv=None):
                         "dir and inject it (LD_PRELOAD) into every "
                         "allocated container; empty = advisory isolation "
                         "only (reference behavior without cGPU)")
+    p.add_argument("--trace-file", default="", metavar="PATH",
+                   help="append a JSONL record per Allocate (ts, units, "
+                        "outcome, per-stage ms) for debugging")
     p.add_argument("--metrics-port", type=int, default=0,
                    help="serve Prometheus /metrics on this port (0 = off)")
     p.add_argument("-v", "--verbose", action="count", default=0)
@@ -150,6 +153,7 @@ def main(argv=None) -> int:
             use_informer=not args.no_informer,
             memguard_path=memguard_path,
             allow_oversize_inventory=args.allow_oversize_inventory,
+            trace_file=args.trace_file,
         ),
     )
     mgr.install_signal_handlers()

@@ -51,6 +51,7 @@ class ManagerOptions:
     memguard_path: str = ""      # host path of libgpushare_memguard.so;
                                  # "" disables VRAM budget enforcement
     allow_oversize_inventory: bool = False
+    trace_file: str = ""         # JSONL per-Allocate trace (debugging)
 
 
 class SharedGPUManager:
@@ -105,6 +106,7 @@ class SharedGPUManager:
             inject_devices=self.opt.inject_devices,
             event_recorder=EventRecorder(self.kube, self.node_name),
             memguard_path=self.opt.memguard_path,
+            trace_file=self.opt.trace_file,
         )
         return GPUSharePlugin(
             gpus,

@@ -198,3 +198,20 @@ def test_memguard_injection():
     c2 = alloc2.allocate(_request([4])).container_responses[0]
     assert consts.ENV_MEMGUARD_LIMIT not in c2.envs
     assert len(c2.mounts) == 0
+
+
+def test_allocate_trace_file(tmp_path):
+    import json as _json
+
+    kube = FakeKubeClient(node_name="node-a")
+    pm = PodManager(kube, "node-a", kubelet_client=kube.as_kubelet(),
+                    cache_ttl=0.0, kubelet_retries=0, apiserver_retries=0)
+    gpus = MockSource.from_spec("1x16GiB").devices()
+    trace = tmp_path / "alloc.jsonl"
+    alloc = Allocator(gpus, pm, trace_file=str(trace))
+    alloc.allocate(_request([4]))
+    alloc.allocate(_request([2]))
+    lines = [_json.loads(l) for l in trace.read_text().splitlines()]
+    assert len(lines) == 2
+    assert lines[0]["ok"] is True and lines[0]["req_units"] == 4
+    assert lines[0]["total_ms"] >= 0

@@ -142,3 +142,44 @@ def test_concurrent_allocates_distinct_pods(harness):
             ]
             == "true"
         )
+
+
+def test_pre_start_container_rpc(tmp_socket_dir):
+    """PreStartContainer: empty-response stub, options say not required
+    (reference parity: server.go:191-193; GetDevicePluginOptions empty)."""
+    from gpushare_amd import consts
+    from gpushare_amd.allocator import Allocator
+    from gpushare_amd.cluster.kubeclient import FakeKubeClient
+    from gpushare_amd.cluster.podmanager import PodManager
+    from gpushare_amd.device.mock_source import MockSource
+    from gpushare_amd.deviceplugin import v1beta1 as api
+    from gpushare_amd.deviceplugin.server import GPUSharePlugin
+    from gpushare_amd.deviceplugin.stubkubelet import StubKubelet
+
+    import grpc
+
+    kube = FakeKubeClient("node-a")
+    gpus = MockSource.from_spec("1x8GiB").devices()
+    pm = PodManager(kube, "node-a", kubelet_client=kube.as_kubelet(),
+                    cache_ttl=0.0)
+    plugin = GPUSharePlugin(gpus, Allocator(gpus, pm),
+                            socket_dir=tmp_socket_dir)
+    kubelet = StubKubelet(tmp_socket_dir)
+    kubelet.start()
+    try:
+        plugin.serve()
+        client = kubelet.wait_for_plugin(consts.RESOURCE_NAME)
+        assert client.options.pre_start_required is False
+        with grpc.insecure_channel(f"unix://{plugin.socket_path}") as ch:
+            pre_start = ch.unary_unary(
+                api.METHOD_PRE_START,
+                request_serializer=lambda m: m.SerializeToString(),
+                response_deserializer=api.PreStartContainerResponse.FromString,
+            )
+            req = api.PreStartContainerRequest()
+            req.devicesIDs.append(f"{gpus[0].uuid}-_-0")
+            resp = pre_start(req, timeout=5)
+            assert resp is not None
+    finally:
+        plugin.stop()
+        kubelet.stop()
