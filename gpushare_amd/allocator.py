@@ -354,6 +354,14 @@ class Allocator:
                     spec.permissions = "rw"
         return responses
 
+    def close(self) -> None:
+        if self._trace is not None:
+            try:
+                self._trace.close()
+            except OSError:
+                pass
+            self._trace = None
+
     def _err_response(self, request, req_units: int) -> "api.AllocateResponse":
         """Poisoned-env failure response (reference: buildErrResponse,
         allocate.go:24-39 — same string format, AMD env names)."""

@@ -250,6 +250,8 @@ class GPUSharePlugin:
                 os.unlink(self.socket_path)
             except OSError:
                 pass
+        if hasattr(self.allocator, "close"):
+            self.allocator.close()
 
     # convenience for tests
     def wait_stopped(self, timeout: float = 5.0) -> None:
