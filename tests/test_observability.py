@@ -80,13 +80,21 @@ def test_metrics_observed_through_allocate():
 
 @pytest.mark.skipif(not metrics.AVAILABLE, reason="prometheus_client missing")
 def test_metrics_http_endpoint():
-    import httpx
+    from gpushare_amd.cluster.httpconn import HttpSession
 
     server = metrics.serve(0)
     try:
         port = server.server_port
-        body = httpx.get(f"http://127.0.0.1:{port}/metrics").text
+        metrics.observe_informer_state(True)
+        metrics.observe_vram_usage({0: 123})
+        s = HttpSession(f"http://127.0.0.1:{port}")
+        status, raw = s.request("GET", "/metrics")
+        body = raw.decode()
+        s.close()
+        assert status == 200
         assert "gpushare_allocate_total" in body
         assert "gpushare_fake_devices" in body
+        assert "gpushare_informer_synced 1.0" in body
+        assert 'gpushare_vram_used_bytes{gpu="0"} 123.0' in body
     finally:
         server.shutdown()
