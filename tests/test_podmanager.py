@@ -144,3 +144,37 @@ def test_real_kubelet_client_over_http():
         kc.close()
     finally:
         server.stop()
+
+
+def test_podgetter_cli_over_tls(tmp_path, capsys):
+    """gpushare-podgetter against a TLS kubelet endpoint (the production
+    shape: self-signed serving cert, bearer token, verification off)."""
+    import json
+    import ssl
+    import subprocess
+
+    from gpushare_amd.cli import podgetter
+    from gpushare_amd.cluster.fasthttp import FastHTTPServer
+
+    cert, key = tmp_path / "t.crt", tmp_path / "t.key"
+    subprocess.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+         "-days", "1", "-subj", "/CN=127.0.0.1",
+         "-keyout", str(key), "-out", str(cert)],
+        check=True, capture_output=True,
+    )
+    ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+    ctx.load_cert_chain(str(cert), str(key))
+    podlist = {"kind": "PodList", "items": []}
+
+    srv = FastHTTPServer(
+        lambda m, p, b: (200, json.dumps(podlist).encode()), ssl_context=ctx
+    ).start()
+    try:
+        rc = podgetter.main(
+            ["--kubelet-port", str(srv.port), "--token", "tok"]
+        )
+        assert rc == 0
+        assert json.loads(capsys.readouterr().out) == podlist
+    finally:
+        srv.stop()
