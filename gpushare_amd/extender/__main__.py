@@ -56,6 +56,12 @@ def main(argv=None) -> int:
     p = argparse.ArgumentParser(prog="gpushare-scheduler-extender")
     p.add_argument("--port", type=int, default=32766)
     p.add_argument("--resync-interval", type=float, default=30.0)
+    p.add_argument("--policy", choices=("binpack", "spread"),
+                   default="binpack",
+                   help="binpack: co-locate tightly (reference behavior); "
+                        "spread: most-free GPU first — trades packing for "
+                        "less interference (measured trade: "
+                        "profiles/colocation_fairness_gpu_box.md)")
     p.add_argument("--no-watch", action="store_true",
                    help="disable the pod watch (auto-release of deleted "
                         "pods' reservations); rely on resync only")
@@ -64,7 +70,8 @@ def main(argv=None) -> int:
     logging.basicConfig(level=logging.INFO, stream=sys.stderr)
 
     kube = RestKubeClient(base_url=args.api_url) if args.api_url else RestKubeClient()
-    extender = GPUShareExtender(kube, resync_interval=args.resync_interval)
+    extender = GPUShareExtender(kube, resync_interval=args.resync_interval,
+                                policy=args.policy)
     # initial discovery retries: the apiserver may lag the daemon at boot
     n = 0
     for attempt in range(30):

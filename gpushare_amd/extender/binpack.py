@@ -42,11 +42,20 @@ class NodeGPUState:
     def free(self, idx: int) -> int:
         return self.per_gpu_units[idx] - self.allocated[idx]
 
-    def best_fit(self, request: int) -> Optional[int]:
+    def best_fit(self, request: int, spread: bool = False) -> Optional[int]:
+        """binpack (default): least feasible free — co-locate tightly.
+        spread: MOST free — isolate tenants (trades packing for less
+        interference; see profiles/colocation_fairness_gpu_box.md for the
+        measured trade)."""
         best, best_free = None, None
         for i in range(len(self.per_gpu_units)):
             f = self.free(i)
-            if f >= request and (best_free is None or f < best_free):
+            if f < request:
+                continue
+            if (
+                best_free is None
+                or (f > best_free if spread else f < best_free)
+            ):
                 best, best_free = i, f
         return best
 
@@ -54,10 +63,12 @@ class NodeGPUState:
         s = set(idxs)
         return sum(1 for i in idxs for p in self.xgmi[i] if p in s) // 2
 
-    def best_fit_multi(self, request: int) -> Optional[dict[int, int]]:
+    def best_fit_multi(
+        self, request: int, spread: bool = False
+    ) -> Optional[dict[int, int]]:
         """Placement map {gpu_idx: units}.  Single GPU when possible;
         otherwise the smallest, most-xGMI-connected, tightest set."""
-        idx = self.best_fit(request)
+        idx = self.best_fit(request, spread=spread)
         if idx is not None:
             return {idx: request}
         n = len(self.per_gpu_units)
@@ -125,7 +136,8 @@ class BinpackState:
             return out
 
     def score_nodes(
-        self, request: int, candidates: list[str], max_score: int = 10
+        self, request: int, candidates: list[str], max_score: int = 10,
+        spread: bool = False,
     ) -> dict[str, int]:
         """Binpack node scoring for the scheduler `prioritize` webhook:
         higher for the node whose placement leaves the least slack — pods
@@ -152,10 +164,11 @@ class BinpackState:
                 if worst == best:
                     scores[name] = max_score
                 else:
-                    # tightest fit -> max_score, loosest feasible -> 1
-                    scores[name] = 1 + round(
-                        (max_score - 1) * (worst - slack) / (worst - best)
-                    )
+                    # binpack: tightest fit -> max_score; spread: loosest
+                    frac = (worst - slack) / (worst - best)
+                    if spread:
+                        frac = 1.0 - frac
+                    scores[name] = 1 + round((max_score - 1) * frac)
             return scores
 
     def assume(self, node: str, request: int) -> Optional[int]:
@@ -171,14 +184,15 @@ class BinpackState:
             st.allocated[idx] += request
             return idx
 
-    def assume_multi(self, node: str, request: int) -> Optional[dict[int, int]]:
+    def assume_multi(self, node: str, request: int,
+                     spread: bool = False) -> Optional[dict[int, int]]:
         """Reserve `request` units across one or more xGMI-adjacent GPUs;
         returns the placement map {gpu_idx: units}."""
         with self._lock:
             st = self.nodes.get(node)
             if st is None:
                 return None
-            split = st.best_fit_multi(request)
+            split = st.best_fit_multi(request, spread=spread)
             if split is None:
                 return None
             for idx, units in split.items():

@@ -23,8 +23,12 @@ log = logging.getLogger(__name__)
 
 
 class GPUShareExtender:
-    def __init__(self, kube_client, resync_interval: float = 30.0):
+    def __init__(self, kube_client, resync_interval: float = 30.0,
+                 policy: str = "binpack"):
+        if policy not in ("binpack", "spread"):
+            raise ValueError(f"unknown placement policy {policy!r}")
         self.kube = kube_client
+        self.policy = policy
         self.state = BinpackState()
         self._lock = threading.Lock()
         self.resync_interval = resync_interval
@@ -130,7 +134,9 @@ class GPUShareExtender:
         request = podutils.gpu_memory_of_pod(pod)
         if request <= 0:
             return None
-        split = self.state.assume_multi(node, request)
+        split = self.state.assume_multi(
+            node, request, spread=(self.policy == "spread")
+        )
         if split is None:
             self.rejected += 1
             return None
@@ -223,7 +229,9 @@ class GPUShareExtender:
         request = podutils.gpu_memory_of_pod(pod)
         if request <= 0:
             return [{"Host": n, "Score": 0} for n in node_names]
-        scores = self.state.score_nodes(request, node_names)
+        scores = self.state.score_nodes(
+            request, node_names, spread=(self.policy == "spread")
+        )
         return [
             {"Host": n, "Score": scores.get(n, 0)} for n in node_names
         ]

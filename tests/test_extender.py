@@ -245,3 +245,42 @@ class TestPrioritize:
             client.close()
         finally:
             server.stop()
+
+
+class TestSpreadPolicy:
+    def test_spread_picks_most_free_gpu(self):
+        from gpushare_amd.cluster.kubeclient import FakeKubeClient
+        from gpushare_amd.extender.core import GPUShareExtender
+
+        kube = FakeKubeClient(node_name="a")
+        ext = GPUShareExtender(kube, resync_interval=3600, policy="spread")
+        ext.register_node("a", [16, 16])
+        p1 = make_pod("p1", mem=4, node="a")
+        del p1["metadata"]["annotations"]
+        kube.add_pod(p1)
+        assert ext.assume(p1, "a") == 0
+        p2 = make_pod("p2", mem=4, node="a")
+        del p2["metadata"]["annotations"]
+        kube.add_pod(p2)
+        # binpack would co-locate on GPU 0; spread must pick GPU 1
+        assert ext.assume(p2, "a") == 1
+
+    def test_spread_scores_prefer_emptier_node(self):
+        from gpushare_amd.cluster.kubeclient import FakeKubeClient
+        from gpushare_amd.extender.core import GPUShareExtender
+
+        kube = FakeKubeClient(node_name="a")
+        ext = GPUShareExtender(kube, resync_interval=3600, policy="spread")
+        ext.register_node("a", [16])
+        ext.register_node("b", [16])
+        ext.state.assume("a", 8)
+        pod = make_pod("p", mem=4, node="")
+        scores = {e["Host"]: e["Score"] for e in ext.prioritize(pod, ["a", "b"])}
+        assert scores["b"] > scores["a"]
+
+    def test_unknown_policy_rejected(self):
+        from gpushare_amd.cluster.kubeclient import FakeKubeClient
+        from gpushare_amd.extender.core import GPUShareExtender
+
+        with pytest.raises(ValueError):
+            GPUShareExtender(FakeKubeClient("a"), policy="nope")
