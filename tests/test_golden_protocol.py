@@ -1,0 +1,557 @@
+"""Wire-compat against the *protoc-generated* kubelet descriptor (golden oracle).
+
+Round-1 gap: `tests/test_protocol.py` oracles the hand-built descriptor in
+`gpushare_amd/deviceplugin/v1beta1.py` against the same protobuf runtime that
+built it — self-referential.  This module closes that gap without needing a
+kubelet binary:
+
+- `tests/fixtures/k8s_deviceplugin_v1beta1.fdp.bin` is the FileDescriptorProto
+  that **protoc produced for upstream Kubernetes** (extracted from the gzipped
+  descriptor embedded in the vendored generated Go,
+  vendor/k8s.io/kubernetes/pkg/kubelet/apis/deviceplugin/v1beta1/api.pb.go;
+  contract source api.proto:23-161).  Message classes built from it share no
+  code with v1beta1.py.
+- A third, even more independent check hand-encodes wire bytes from the
+  protobuf wire-format spec directly (varints + tags written in the test), so
+  even the protobuf runtime is not a shared oracle for the core messages.
+- A live gRPC interop test talks to our plugin server using ONLY
+  golden-descriptor clients (method paths derived from the golden service
+  descriptors), covering Register, GetDevicePluginOptions, ListAndWatch and
+  Allocate end-to-end over a unix socket.
+- Unknown-field tolerance: a modern kubelet sends `Device.topology` (field 3,
+  added after the vendored snapshot); our decoder must tolerate and our
+  encoder's output must parse in a decoder that knows field 3.
+"""
+
+from __future__ import annotations
+
+import os
+import threading
+from concurrent import futures
+
+import grpc
+import pytest
+from google.protobuf import descriptor_pb2, descriptor_pool, message_factory
+
+from gpushare_amd import consts
+from gpushare_amd.allocator import Allocator
+from gpushare_amd.cluster.kubeclient import FakeKubeClient
+from gpushare_amd.cluster.podmanager import PodManager
+from gpushare_amd.device.mock_source import MockSource
+from gpushare_amd.deviceplugin import v1beta1 as ours
+from gpushare_amd.deviceplugin.server import GPUSharePlugin
+
+from helpers import make_pod
+
+FIXTURE = os.path.join(
+    os.path.dirname(__file__), "fixtures", "k8s_deviceplugin_v1beta1.fdp.bin"
+)
+
+
+# --------------------------------------------------------------------------- #
+# golden descriptor loading
+# --------------------------------------------------------------------------- #
+def _load_golden_fdp() -> descriptor_pb2.FileDescriptorProto:
+    fdp = descriptor_pb2.FileDescriptorProto()
+    with open(FIXTURE, "rb") as fh:
+        fdp.ParseFromString(fh.read())
+    return fdp
+
+
+def _scrub(msg_proto: descriptor_pb2.DescriptorProto) -> None:
+    """Drop gogoproto extension options (Go-codegen hints only; they do not
+    affect the wire format) so the descriptor loads without gogo.proto.
+    map_entry is semantic and is preserved."""
+    for field in msg_proto.field:
+        field.ClearField("options")
+    for nested in msg_proto.nested_type:
+        map_entry = nested.options.map_entry
+        nested.ClearField("options")
+        if map_entry:
+            nested.options.map_entry = True
+        _scrub(nested)
+    map_entry = msg_proto.options.map_entry
+    msg_proto.ClearField("options")
+    if map_entry:
+        msg_proto.options.map_entry = True
+
+
+@pytest.fixture(scope="module")
+def golden():
+    """(file_descriptor_proto, {message_name: dynamic class}, services)."""
+    fdp = _load_golden_fdp()
+    loadable = descriptor_pb2.FileDescriptorProto()
+    loadable.CopyFrom(fdp)
+    # gogo dependency + file/service options are codegen metadata, not wire.
+    del loadable.dependency[:]
+    loadable.ClearField("options")
+    for svc in loadable.service:
+        svc.ClearField("options")
+        for method in svc.method:
+            method.ClearField("options")
+    for mt in loadable.message_type:
+        _scrub(mt)
+    loadable.name = "golden/kubelet_v1beta1_api.proto"
+
+    pool = descriptor_pool.DescriptorPool()
+    pool.Add(loadable)
+    classes = {
+        mt.name: message_factory.GetMessageClass(
+            pool.FindMessageTypeByName(f"{fdp.package}.{mt.name}")
+        )
+        for mt in fdp.message_type
+    }
+    return fdp, classes
+
+
+# --------------------------------------------------------------------------- #
+# 1. structural parity: our hand-built descriptor vs protoc's
+# --------------------------------------------------------------------------- #
+def _field_shape(f: descriptor_pb2.FieldDescriptorProto):
+    type_name = f.type_name.rsplit(".", 1)[-1] if f.type_name else ""
+    return (f.name, f.number, int(f.type), int(f.label), type_name)
+
+
+def _message_shapes(messages, prefix=""):
+    out = {}
+    for mt in messages:
+        name = f"{prefix}{mt.name}"
+        out[name] = sorted(_field_shape(f) for f in mt.field)
+        out.update(_message_shapes(mt.nested_type, prefix=f"{name}."))
+    return out
+
+
+def test_structure_matches_protoc_descriptor(golden):
+    fdp, _ = golden
+    golden_shapes = _message_shapes(fdp.message_type)
+    # v1beta1._build_file() is the module's own construction path
+    our_shapes = _message_shapes(ours._build_file().message_type)
+    assert our_shapes == golden_shapes
+
+
+def test_package_and_service_names_match(golden):
+    fdp, _ = golden
+    assert fdp.package == ours._PKG
+    services = {s.name: s for s in fdp.service}
+    assert set(services) == {"Registration", "DevicePlugin"}
+
+    reg = services["Registration"]
+    assert [(m.name, m.server_streaming) for m in reg.method] == [
+        ("Register", False)
+    ]
+    dp = services["DevicePlugin"]
+    assert [(m.name, m.server_streaming) for m in dp.method] == [
+        ("GetDevicePluginOptions", False),
+        ("ListAndWatch", True),
+        ("Allocate", False),
+        ("PreStartContainer", False),
+    ]
+
+    # the gRPC paths our server/clients use must be derivable from the
+    # golden descriptor alone
+    assert ours.METHOD_REGISTER == f"/{fdp.package}.Registration/Register"
+    for m in dp.method:
+        path = f"/{fdp.package}.DevicePlugin/{m.name}"
+        assert path in {
+            ours.METHOD_GET_OPTIONS,
+            ours.METHOD_LIST_AND_WATCH,
+            ours.METHOD_ALLOCATE,
+            ours.METHOD_PRE_START,
+        }
+
+
+# --------------------------------------------------------------------------- #
+# 2. byte-identical serialization, both directions
+# --------------------------------------------------------------------------- #
+def _sample_payloads():
+    """(message name, dict of plain-python field values) for every message."""
+    mount = {"container_path": "/c", "host_path": "/h", "read_only": True}
+    devspec = {
+        "container_path": consts.DEV_KFD,
+        "host_path": consts.DEV_KFD,
+        "permissions": "rw",
+    }
+    car = {
+        "envs": {
+            consts.ENV_RESOURCE_INDEX: "3",
+            consts.ENV_RESOURCE_BY_POD: "72",
+            "ROCR_VISIBLE_DEVICES": "3",
+        },
+        "mounts": [mount],
+        "devices": [devspec, dict(devspec, container_path="/dev/dri/renderD128")],
+        "annotations": {"a": "1", "b": "2"},
+    }
+    return [
+        ("DevicePluginOptions", {"pre_start_required": True}),
+        ("DevicePluginOptions", {}),
+        (
+            "RegisterRequest",
+            {
+                "version": consts.API_VERSION,
+                "endpoint": "amdgpushare.sock",
+                "resource_name": consts.RESOURCE_NAME,
+                "options": {"pre_start_required": False},
+            },
+        ),
+        ("Empty", {}),
+        (
+            "ListAndWatchResponse",
+            {
+                "devices": [
+                    {"ID": f"gpu-{i}-_-{j}", "health": consts.HEALTHY}
+                    for i in range(4)
+                    for j in range(3)
+                ]
+            },
+        ),
+        ("Device", {"ID": "mock-00-_-1", "health": "Unhealthy"}),
+        ("Device", {"ID": "", "health": ""}),  # proto3 default elision
+        ("PreStartContainerRequest", {"devicesIDs": ["a", "b", "c"]}),
+        ("PreStartContainerResponse", {}),
+        (
+            "AllocateRequest",
+            {
+                "container_requests": [
+                    {"devicesIDs": ["x-_-0", "x-_-1"]},
+                    {"devicesIDs": []},
+                ]
+            },
+        ),
+        ("AllocateResponse", {"container_responses": [car, {}]}),
+        ("Mount", mount),
+        ("DeviceSpec", devspec),
+    ]
+
+
+def _fill(msg, values):
+    for key, val in values.items():
+        field = msg.DESCRIPTOR.fields_by_name[key]
+        if field.message_type is not None and field.message_type.GetOptions().map_entry:
+            getattr(msg, key).update(val)
+        elif field.is_repeated:
+            target = getattr(msg, key)
+            for item in val:
+                if isinstance(item, dict):
+                    _fill(target.add(), item)
+                else:
+                    target.append(item)
+        elif isinstance(val, dict):
+            _fill(getattr(msg, key), val)
+        else:
+            setattr(msg, key, val)
+    return msg
+
+
+@pytest.mark.parametrize("name,values", _sample_payloads())
+def test_serialization_byte_identical(golden, name, values):
+    _, classes = golden
+    theirs = _fill(classes[name](), values).SerializeToString(deterministic=True)
+    mine = _fill(getattr(ours, name)(), values).SerializeToString(
+        deterministic=True
+    )
+    assert mine == theirs
+
+
+@pytest.mark.parametrize("name,values", _sample_payloads())
+def test_cross_parse_roundtrip(golden, name, values):
+    _, classes = golden
+    # our bytes parse in the protoc-derived class and re-serialize identically
+    mine = _fill(getattr(ours, name)(), values).SerializeToString(
+        deterministic=True
+    )
+    parsed = classes[name]()
+    parsed.ParseFromString(mine)
+    assert parsed.SerializeToString(deterministic=True) == mine
+    # and the reverse
+    theirs = _fill(classes[name](), values).SerializeToString(deterministic=True)
+    back = getattr(ours, name)()
+    back.ParseFromString(theirs)
+    assert back.SerializeToString(deterministic=True) == theirs
+
+
+# --------------------------------------------------------------------------- #
+# 3. first-principles wire bytes (no protobuf runtime involved)
+# --------------------------------------------------------------------------- #
+def _varint(n: int) -> bytes:
+    out = bytearray()
+    while True:
+        bits = n & 0x7F
+        n >>= 7
+        if n:
+            out.append(bits | 0x80)
+        else:
+            out.append(bits)
+            return bytes(out)
+
+
+def _tag(field_number: int, wire_type: int) -> bytes:
+    return _varint((field_number << 3) | wire_type)
+
+
+def _ld(field_number: int, payload: bytes) -> bytes:
+    """Length-delimited field (wire type 2)."""
+    return _tag(field_number, 2) + _varint(len(payload)) + payload
+
+
+def _s(field_number: int, text: str) -> bytes:
+    return _ld(field_number, text.encode())
+
+
+def test_register_request_wire_bytes_first_principles():
+    expected = (
+        _s(1, "v1beta1")
+        + _s(2, "amdgpushare.sock")
+        + _s(3, "aliyun.com/gpu-mem")
+    )
+    msg = ours.RegisterRequest(
+        version="v1beta1",
+        endpoint="amdgpushare.sock",
+        resource_name="aliyun.com/gpu-mem",
+    )
+    assert msg.SerializeToString(deterministic=True) == expected
+
+
+def test_list_and_watch_wire_bytes_first_principles():
+    device = _s(1, "amd-0-_-7") + _s(2, "Healthy")
+    expected = _ld(1, device) * 1  # repeated field 1
+    resp = ours.ListAndWatchResponse(
+        devices=[ours.Device(ID="amd-0-_-7", health="Healthy")]
+    )
+    assert resp.SerializeToString(deterministic=True) == expected
+
+
+def test_allocate_response_wire_bytes_first_principles():
+    # envs map entry = nested message {key:1, value:2}; map fields serialize
+    # as repeated entries, deterministic order = sorted by key
+    entry1 = _s(1, "A") + _s(2, "1")
+    entry2 = _s(1, "B") + _s(2, "2")
+    devspec = _s(1, "/dev/kfd") + _s(2, "/dev/kfd") + _s(3, "rw")
+    car = _ld(1, entry1) + _ld(1, entry2) + _ld(3, devspec)
+    expected = _ld(1, car)
+    resp = ours.AllocateResponse()
+    cr = resp.container_responses.add()
+    cr.envs["B"] = "2"
+    cr.envs["A"] = "1"
+    dev = cr.devices.add()
+    dev.container_path = "/dev/kfd"
+    dev.host_path = "/dev/kfd"
+    dev.permissions = "rw"
+    assert resp.SerializeToString(deterministic=True) == expected
+
+
+def test_bool_and_default_elision_wire_bytes():
+    # proto3: false bool and empty strings are omitted entirely
+    assert ours.DevicePluginOptions(
+        pre_start_required=False
+    ).SerializeToString() == b""
+    assert (
+        ours.DevicePluginOptions(pre_start_required=True).SerializeToString()
+        == _tag(1, 0) + b"\x01"
+    )
+    assert ours.Device(ID="", health="").SerializeToString() == b""
+
+
+# --------------------------------------------------------------------------- #
+# 4. unknown-field tolerance (modern kubelet: Device.topology = field 3)
+# --------------------------------------------------------------------------- #
+def _device_with_topology(dev_id: str, health: str) -> bytes:
+    # TopologyInfo{ repeated NUMANode nodes = 1 }; NUMANode{ int64 ID = 1 }
+    numa_node = _tag(1, 0) + _varint(2)          # ID = 2
+    topology = _ld(1, numa_node)
+    return _s(1, dev_id) + _s(2, health) + _ld(3, topology)
+
+
+def test_device_tolerates_future_topology_field():
+    raw = _device_with_topology("amd-1-_-4", "Healthy")
+    dev = ours.Device()
+    dev.ParseFromString(raw)
+    assert dev.ID == "amd-1-_-4"
+    assert dev.health == "Healthy"
+    # unknown fields must be preserved on re-serialization (kubelet may
+    # reflect devices back through PreStartContainer paths)
+    assert dev.SerializeToString() == raw
+
+
+def test_list_and_watch_tolerates_future_topology_field():
+    payload = _ld(1, _device_with_topology("a-_-0", "Healthy")) + _ld(
+        1, _device_with_topology("a-_-1", "Unhealthy")
+    )
+    resp = ours.ListAndWatchResponse()
+    resp.ParseFromString(payload)
+    assert [(d.ID, d.health) for d in resp.devices] == [
+        ("a-_-0", "Healthy"),
+        ("a-_-1", "Unhealthy"),
+    ]
+
+
+def test_register_request_with_unknown_field_5():
+    # future RegisterRequest additions must not break a registration server
+    raw = ours.RegisterRequest(
+        version="v1beta1", endpoint="e.sock", resource_name="r"
+    ).SerializeToString() + _s(5, "future")
+    req = ours.RegisterRequest()
+    req.ParseFromString(raw)
+    assert req.version == "v1beta1"
+    assert req.resource_name == "r"
+
+
+# --------------------------------------------------------------------------- #
+# 5. live gRPC interop: golden-descriptor client <-> our plugin server
+# --------------------------------------------------------------------------- #
+class _GoldenKubelet:
+    """Registration server + DevicePlugin client built ONLY from the golden
+    descriptor: method paths from the golden service descriptors, serializers
+    from the protoc-derived classes.  Stands in for the real kubelet's
+    grpc-go endpoint as closely as the sandbox allows."""
+
+    def __init__(self, fdp, classes, socket_dir: str):
+        self.fdp = fdp
+        self.classes = classes
+        self.socket_dir = socket_dir
+        self.socket_path = os.path.join(socket_dir, consts.KUBELET_SOCKET_NAME)
+        self.registered = threading.Event()
+        self.register_request = None
+        self._server = None
+
+    def _path(self, service: str, method: str) -> str:
+        svc = next(s for s in self.fdp.service if s.name == service)
+        m = next(mm for mm in svc.method if mm.name == method)
+        return f"/{self.fdp.package}.{service}/{m.name}"
+
+    def start(self):
+        def register(request_bytes, context):
+            req = self.classes["RegisterRequest"]()
+            req.ParseFromString(request_bytes)
+            self.register_request = req
+            self.registered.set()
+            return self.classes["Empty"]().SerializeToString()
+
+        handler = grpc.method_handlers_generic_handler(
+            f"{self.fdp.package}.Registration",
+            {
+                "Register": grpc.unary_unary_rpc_method_handler(
+                    register,
+                    request_deserializer=None,   # raw bytes in
+                    response_serializer=None,    # raw bytes out
+                )
+            },
+        )
+        self._server = grpc.server(futures.ThreadPoolExecutor(max_workers=4))
+        self._server.add_generic_rpc_handlers((handler,))
+        self._server.add_insecure_port(f"unix://{self.socket_path}")
+        self._server.start()
+
+    def stop(self):
+        if self._server:
+            self._server.stop(grace=0.5)
+
+    def dial_plugin(self, endpoint: str):
+        channel = grpc.insecure_channel(
+            f"unix://{os.path.join(self.socket_dir, endpoint)}"
+        )
+        grpc.channel_ready_future(channel).result(timeout=5)
+        cls = self.classes
+        calls = {
+            "options": channel.unary_unary(
+                self._path("DevicePlugin", "GetDevicePluginOptions"),
+                request_serializer=lambda m: m.SerializeToString(),
+                response_deserializer=cls["DevicePluginOptions"].FromString,
+            ),
+            "law": channel.unary_stream(
+                self._path("DevicePlugin", "ListAndWatch"),
+                request_serializer=lambda m: m.SerializeToString(),
+                response_deserializer=cls["ListAndWatchResponse"].FromString,
+            ),
+            "allocate": channel.unary_unary(
+                self._path("DevicePlugin", "Allocate"),
+                request_serializer=lambda m: m.SerializeToString(),
+                response_deserializer=cls["AllocateResponse"].FromString,
+            ),
+            "prestart": channel.unary_unary(
+                self._path("DevicePlugin", "PreStartContainer"),
+                request_serializer=lambda m: m.SerializeToString(),
+                response_deserializer=cls[
+                    "PreStartContainerResponse"
+                ].FromString,
+            ),
+        }
+        return channel, calls
+
+
+@pytest.fixture
+def golden_harness(golden, tmp_socket_dir):
+    fdp, classes = golden
+    kubelet = _GoldenKubelet(fdp, classes, tmp_socket_dir)
+    kubelet.start()
+
+    kube = FakeKubeClient(node_name="node-a")
+    pm = PodManager(
+        kube,
+        "node-a",
+        kubelet_client=kube.as_kubelet(),
+        cache_ttl=0.0,
+        kubelet_retries=0,
+        kubelet_retry_interval=0.0,
+        apiserver_retries=0,
+        apiserver_retry_interval=0.0,
+    )
+    gpus = MockSource.from_spec("2x8GiB").devices()
+    plugin = GPUSharePlugin(gpus, Allocator(gpus, pm), socket_dir=tmp_socket_dir)
+    plugin.serve()  # start() + register() against the golden kubelet
+
+    yield kubelet, kube, plugin
+    plugin.stop()
+    kubelet.stop()
+
+
+def test_golden_client_full_flow(golden_harness):
+    kubelet, kube, plugin = golden_harness
+    _, classes = (None, kubelet.classes)
+
+    # --- registration arrived and decoded with protoc-derived classes
+    assert kubelet.registered.wait(timeout=5)
+    req = kubelet.register_request
+    assert req.version == consts.API_VERSION
+    assert req.resource_name == consts.RESOURCE_NAME
+    assert req.endpoint == consts.SERVER_SOCK_NAME
+
+    channel, calls = kubelet.dial_plugin(req.endpoint)
+    try:
+        # --- GetDevicePluginOptions
+        opts = calls["options"](classes["Empty"](), timeout=5)
+        assert opts.pre_start_required is False
+
+        # --- ListAndWatch first payload (pre-encoded native codec bytes
+        #     must decode in the protoc-derived class)
+        stream = calls["law"](classes["Empty"]())
+        first = next(stream)
+        devices = {d.ID: d.health for d in first.devices}
+        assert len(devices) == 16  # 2 GPUs x 8 GiB grains
+        assert set(devices.values()) == {consts.HEALTHY}
+
+        # --- health flip: resent list flips one GPU's grains
+        plugin.set_gpu_health(1, healthy=False)
+        second = next(stream)
+        unhealthy = [d.ID for d in second.devices if d.health != consts.HEALTHY]
+        assert len(unhealthy) == 8
+        stream.cancel()
+
+        # --- Allocate round-trip with golden request/response classes
+        kube.add_pod(make_pod("golden-pod", 4, gpu_idx=0))
+        areq = classes["AllocateRequest"]()
+        ids = sorted(i for i in devices if i.split("-_-")[0].endswith("00"))[:4]
+        areq.container_requests.add().devicesIDs.extend(ids)
+        resp = calls["allocate"](areq, timeout=5)
+        assert len(resp.container_responses) == 1
+        envs = resp.container_responses[0].envs
+        assert envs[consts.ENV_RESOURCE_INDEX] == "0"
+        assert envs[consts.ENV_RESOURCE_BY_POD] == "4"
+        host_paths = {d.host_path for d in resp.container_responses[0].devices}
+        assert consts.DEV_KFD in host_paths
+
+        # --- PreStartContainer
+        psr = classes["PreStartContainerRequest"]()
+        psr.devicesIDs.extend(ids)
+        calls["prestart"](psr, timeout=5)
+    finally:
+        channel.close()
