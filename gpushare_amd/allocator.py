@@ -28,6 +28,7 @@ revalidates with one forced refresh on a miss, instead of the reference's
 from __future__ import annotations
 
 import logging
+import os
 import threading
 import time
 from typing import Optional
@@ -107,8 +108,14 @@ class Allocator:
         self.stats = AllocateStats()
         # optional JSONL trace of every Allocate (the
         # reference has no tracing although Allocate latency is the
-        # north-star metric); one line per call, append-only
-        self._trace = open(trace_file, "a") if trace_file else None
+        # north-star metric); one line per call.  Raw O_APPEND fd, one
+        # os.write per line: concurrent gRPC workers (max_workers=64)
+        # would interleave buffered text-mode writes mid-line
+        self._trace = (
+            os.open(trace_file, os.O_WRONLY | os.O_CREAT | os.O_APPEND, 0o644)
+            if trace_file
+            else None
+        )
         # Matching runs under a short in-memory critical section; the
         # ASSIGNED patch happens OUTSIDE it.  A matched pod is "claimed"
         # (uid -> expiry) so concurrent Allocates skip it; claims are
@@ -144,7 +151,11 @@ class Allocator:
                     self._unclaim(uid)
                     return self._err_response(request, req_units)
                 resp = self._build_response(
-                    request, req_units, gpus, pod_uid=uid
+                    request,
+                    req_units,
+                    gpus,
+                    pod_uid=uid,
+                    split=podutils.gpu_split_from_pod(pod),
                 )
                 tp = time.perf_counter()
                 patched = self.pods.mark_assigned(pod)
@@ -176,15 +187,15 @@ class Allocator:
             if self._trace is not None:
                 import json as _json
 
-                self._trace.write(_json.dumps({
+                line = _json.dumps({
                     "ts": time.time(),
                     "req_units": req_units,
                     "ok": ok,
                     "total_ms": round(total * 1e3, 3),
                     "match_ms": round(list_t * 1e3, 3),
                     "patch_ms": round(patch_t * 1e3, 3),
-                }) + "\n")
-                self._trace.flush()
+                }) + "\n"
+                os.write(self._trace, line.encode())
 
     # ------------------------------------------------------------------ #
     def _match_and_claim(self, req_units: int) -> Optional[dict]:
@@ -307,6 +318,7 @@ class Allocator:
         req_units: int,
         gpus: list[PhysicalGPU],
         pod_uid: str = "",
+        split: Optional[dict] = None,
     ) -> "api.AllocateResponse":
         """Container envs + AMD device nodes for the bound GPU(s).  For a
         multi-GPU placement every container sees the whole set (ROCr
@@ -319,8 +331,18 @@ class Allocator:
             g.extras.get("rocr_uuid") or str(g.index) for g in gpus
         )
         hip_visible = ",".join(str(i) for i in range(len(gpus)))
+        # per-GPU sub-budgets (units) still unclaimed by earlier containers:
+        # greedily carve each container's per-device caps out of the
+        # extender's split so caps sum exactly to the split per GPU and to
+        # each container's own total (a pod-total-only budget
+        # let a tenant concentrate everything on one split member)
+        split_remaining = (
+            [split.get(g.index, 0) for g in gpus]
+            if split and len(gpus) > 1
+            else None
+        )
         responses = api.AllocateResponse()
-        for cr in request.container_requests:
+        for i, cr in enumerate(request.container_requests):
             c = responses.container_responses.add()
             c.envs[consts.ENV_ROCR_VISIBLE] = rocr_id
             c.envs[consts.ENV_HIP_VISIBLE] = hip_visible
@@ -332,13 +354,29 @@ class Allocator:
                 c.envs[consts.ENV_CGPU_DISABLE] = "true"
             elif self.memguard_path:
                 shift = 30 if self.unit == consts.GIB else 20
+                container_units = len(cr.devicesIDs)
                 c.envs[consts.ENV_MEMGUARD_LIMIT] = str(
-                    len(cr.devicesIDs) << shift
+                    container_units << shift
                 )
                 c.envs["LD_PRELOAD"] = consts.MEMGUARD_CONTAINER_PATH
                 if pod_uid:
                     # scopes the shared budget counter under /dev/shm
                     c.envs[consts.ENV_MEMGUARD_POD_UID] = pod_uid
+                # container-scoped accounting table (see consts)
+                c.envs[consts.ENV_MEMGUARD_CONTAINER_TOKEN] = f"c{i}"
+                if split_remaining is not None:
+                    # carve this container's per-device caps from what the
+                    # split still has available, visible-ordinal order
+                    need = container_units
+                    caps = []
+                    for j in range(len(gpus)):
+                        take = min(split_remaining[j], need)
+                        caps.append(take << shift)
+                        split_remaining[j] -= take
+                        need -= take
+                    c.envs[consts.ENV_MEMGUARD_PER_DEVICE] = ",".join(
+                        str(b) for b in caps
+                    )
                 m = c.mounts.add()
                 m.container_path = consts.MEMGUARD_CONTAINER_PATH
                 m.host_path = self.memguard_path
@@ -357,7 +395,7 @@ class Allocator:
     def close(self) -> None:
         if self._trace is not None:
             try:
-                self._trace.close()
+                os.close(self._trace)
             except OSError:
                 pass
             self._trace = None

@@ -64,17 +64,23 @@ class TrackedThreadingHTTPServer(ThreadingHTTPServer):
 
 
 def make_ssl_context(verify) -> Optional[ssl.SSLContext]:
-    """httpx-style verify: SSLContext → as-is, str → CA file, falsy → no
-    verification (the kubelet's serving cert is rarely CA-signed; the
-    reference forces insecure too, client.go:75-99)."""
+    """httpx-style verify: SSLContext → as-is, str → CA file, ``False``
+    (explicit!) → no verification, ``None``/``True`` → verified against
+    the system store.  Verification is the DEFAULT: only the
+    kubelet client opts out explicitly — its serving cert is rarely
+    CA-signed and the reference forces insecure there too
+    (client.go:75-99) — while apiserver sessions, including long-lived
+    watch streams, must not silently run unverified."""
     if isinstance(verify, ssl.SSLContext):
         return verify
     if isinstance(verify, str):
         return ssl.create_default_context(cafile=verify)
-    ctx = ssl.create_default_context()
-    ctx.check_hostname = False
-    ctx.verify_mode = ssl.CERT_NONE
-    return ctx
+    if verify is False:
+        ctx = ssl.create_default_context()
+        ctx.check_hostname = False
+        ctx.verify_mode = ssl.CERT_NONE
+        return ctx
+    return ssl.create_default_context()
 
 
 class HttpSession:

@@ -89,14 +89,28 @@ class RestKubeClient:
             user = next(
                 u["user"] for u in cfg["users"] if u["name"] == ctx["user"]
             )
-            verify: object = cluster.get("certificate-authority", False)
+            # verified TLS is the default: a kubeconfig CA —
+            # path or inline certificate-authority-data — pins the
+            # apiserver cert; only an explicit insecure-skip-tls-verify
+            # turns verification off; neither present ⇒ system trust store
+            ca_data = cluster.get("certificate-authority-data")
+            verify: object = cluster.get("certificate-authority", None)
+            if verify is None and ca_data:
+                import base64
+
+                verify = ssl.create_default_context(
+                    cadata=base64.b64decode(ca_data).decode()
+                )
             if cluster.get("insecure-skip-tls-verify"):
                 verify = False
             token = user.get("token")
             if not token and user.get("client-certificate"):
-                sslctx = ssl.create_default_context(
-                    cafile=cluster.get("certificate-authority")
-                )
+                if isinstance(verify, ssl.SSLContext):
+                    sslctx = verify
+                else:
+                    sslctx = ssl.create_default_context(
+                        cafile=verify if isinstance(verify, str) else None
+                    )
                 if verify is False:
                     sslctx.check_hostname = False
                     sslctx.verify_mode = ssl.CERT_NONE
@@ -113,7 +127,9 @@ class RestKubeClient:
         if os.path.exists(token_path):
             token = open(token_path).read().strip()
         ca = os.path.join(SA_DIR, "ca.crt")
-        verify = ca if os.path.exists(ca) else False
+        # no mounted CA ⇒ system trust store (verified) — never silently
+        # insecure for apiserver traffic
+        verify = ca if os.path.exists(ca) else None
         return f"https://{host}:{port}", token, verify
 
     # -- verbs ---------------------------------------------------------------

@@ -83,6 +83,16 @@ DEV_DRI_DIR = "/dev/dri"
 # sets ENV_MEMGUARD_LIMIT to the container's gpu-mem share in bytes.
 ENV_MEMGUARD_LIMIT = "GPUSHARE_MEM_LIMIT_BYTES"
 ENV_MEMGUARD_POD_UID = "GPUSHARE_POD_UID"
+# Per-container token scoping the /dev/shm accounting table: containers of
+# one pod share /dev/shm but not a PID namespace, so liveness sweeps must
+# never see a sibling container's pids.  The kubelet batches
+# every gpu-mem container of a pod into one Allocate call, so the container
+# index within the request is unique per pod.
+ENV_MEMGUARD_CONTAINER_TOKEN = "GPUSHARE_CONTAINER_TOKEN"
+# Comma-separated per-visible-ordinal byte caps (aligned with the injected
+# HIP_VISIBLE_DEVICES order) enforcing the extender's per-GPU split on
+# multi-GPU placements.
+ENV_MEMGUARD_PER_DEVICE = "GPUSHARE_MEM_LIMIT_BYTES_PER_DEVICE"
 MEMGUARD_CONTAINER_PATH = "/usr/local/lib/gpushare/libgpushare_memguard.so"
 
 # Node label opting a node out of kernel-level isolation (analogue of the

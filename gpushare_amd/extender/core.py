@@ -149,22 +149,41 @@ class GPUShareExtender:
             consts.ENV_ASSIGNED_FLAG: "false",
         }
         if len(split) > 1:
-            container = next(
+            # carve the split across the pod's GPU-requesting containers in
+            # spec order (reference per-container map format,
+            # cmd/inspect/nodeinfo.go:244-271): each container's per-GPU
+            # units sum to its own gpu-mem limit, and per GPU the
+            # containers sum to the split — so inspect/top attribution and
+            # the plugin's per-device budget carve agree
+            requesting = [
                 (
-                    c.get("name", "main")
-                    for c in pod.get("spec", {}).get("containers", [])
-                    if int(
+                    c.get("name", f"c{i}"),
+                    int(
                         c.get("resources", {})
                         .get("limits", {})
                         .get(consts.RESOURCE_NAME, 0)
-                    )
-                    > 0
-                ),
-                "main",
-            )
-            anns[consts.ANN_GPUSHARE_ALLOCATION] = json.dumps(
-                {container: {str(i): u for i, u in sorted(split.items())}}
-            )
+                    ),
+                )
+                for i, c in enumerate(pod.get("spec", {}).get("containers", []))
+            ]
+            requesting = [(n, u) for n, u in requesting if u > 0]
+            if not requesting:
+                requesting = [("main", request)]
+            remaining = dict(sorted(split.items()))
+            alloc_map: dict[str, dict[str, int]] = {}
+            for cname, units in requesting:
+                need = units
+                per_gpu: dict[str, int] = {}
+                for gpu_idx in list(remaining):
+                    if need <= 0:
+                        break
+                    take = min(remaining[gpu_idx], need)
+                    if take > 0:
+                        per_gpu[str(gpu_idx)] = take
+                        remaining[gpu_idx] -= take
+                        need -= take
+                alloc_map[cname] = per_gpu
+            anns[consts.ANN_GPUSHARE_ALLOCATION] = json.dumps(alloc_map)
         patch = {"metadata": {"annotations": anns}}
         try:
             self.kube.patch_pod(ns, name, patch, parse=False)

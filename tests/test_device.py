@@ -166,7 +166,95 @@ def test_memguard_lib_loads_and_reads_env(tmp_path):
     out = subprocess.run(
         [sys.executable, "-c", code], env=env, capture_output=True, text=True
     )
-    shm = f"/dev/shm/gpushare.memguard.cputest-{os.getpid()}"
+    shm = f"/dev/shm/gpushare.memguard.cputest-{os.getpid()}.c"
     if os.path.exists(shm):
         os.unlink(shm)
     assert "CPU_MEMGUARD_OK" in out.stdout, out.stderr
+
+
+def test_memguard_container_token_scopes_table():
+    """Containers of one pod share /dev/shm but not a PID namespace, so
+    each container must get its OWN accounting table: the
+    GPUSHARE_CONTAINER_TOKEN env is part of the table path."""
+    import os
+    import subprocess
+    import sys
+
+    import gpushare_amd
+
+    lib = os.path.join(
+        os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+    )
+    if not os.path.exists(lib):
+        import pytest
+
+        pytest.skip("memguard not built")
+    uid = f"toktest-{os.getpid()}"
+    code = (
+        "import ctypes\n"
+        f"l = ctypes.CDLL({lib!r})\n"
+        "l.gpushare_memguard_used.restype = ctypes.c_int64\n"
+        "assert l.gpushare_memguard_used() == 0\n"
+    )
+    paths = []
+    try:
+        for token in ("c0", "c1"):
+            env = dict(os.environ)
+            env["GPUSHARE_MEM_LIMIT_BYTES"] = str(1 << 30)
+            env["GPUSHARE_POD_UID"] = uid
+            env["GPUSHARE_CONTAINER_TOKEN"] = token
+            out = subprocess.run(
+                [sys.executable, "-c", code],
+                env=env,
+                capture_output=True,
+                text=True,
+            )
+            assert out.returncode == 0, out.stderr
+            paths.append(f"/dev/shm/gpushare.memguard.{uid}.{token}")
+        # both containers created distinct tables
+        for p in paths:
+            assert os.path.exists(p), p
+    finally:
+        for p in paths:
+            if os.path.exists(p):
+                os.unlink(p)
+
+
+def test_memguard_per_device_limits_parsed():
+    """GPUSHARE_MEM_LIMIT_BYTES_PER_DEVICE drives per-ordinal caps; absent
+    entries stay uncapped (-1)."""
+    import os
+    import subprocess
+    import sys
+
+    import gpushare_amd
+
+    lib = os.path.join(
+        os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+    )
+    if not os.path.exists(lib):
+        import pytest
+
+        pytest.skip("memguard not built")
+    uid = f"devtest-{os.getpid()}"
+    code = (
+        "import ctypes\n"
+        f"l = ctypes.CDLL({lib!r})\n"
+        "l.gpushare_memguard_dev_limit.restype = ctypes.c_int64\n"
+        "l.gpushare_memguard_dev_limit.argtypes = [ctypes.c_int]\n"
+        "assert l.gpushare_memguard_dev_limit(0) == 1 << 30, l.gpushare_memguard_dev_limit(0)\n"
+        "assert l.gpushare_memguard_dev_limit(1) == 2 << 30\n"
+        "assert l.gpushare_memguard_dev_limit(2) == -1\n"
+        "print('PER_DEVICE_OK')\n"
+    )
+    env = dict(os.environ)
+    env["GPUSHARE_MEM_LIMIT_BYTES"] = str(3 << 30)
+    env["GPUSHARE_MEM_LIMIT_BYTES_PER_DEVICE"] = f"{1 << 30},{2 << 30}"
+    env["GPUSHARE_POD_UID"] = uid
+    out = subprocess.run(
+        [sys.executable, "-c", code], env=env, capture_output=True, text=True
+    )
+    shm = f"/dev/shm/gpushare.memguard.{uid}.c"
+    if os.path.exists(shm):
+        os.unlink(shm)
+    assert "PER_DEVICE_OK" in out.stdout, out.stderr

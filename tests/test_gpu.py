@@ -324,7 +324,7 @@ def test_memguard_budget_shared_across_processes():
         os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
     )
     pod_uid = f"test-{uuid_mod.uuid4().hex[:12]}"
-    shm_path = f"/dev/shm/gpushare.memguard.{pod_uid}"
+    shm_path = f"/dev/shm/gpushare.memguard.{pod_uid}.c"
     env = dict(os.environ)
     env["LD_PRELOAD"] = lib
     env["GPUSHARE_MEM_LIMIT_BYTES"] = str(2 << 30)
@@ -410,7 +410,7 @@ def test_eight_tenants_shared_gpu_with_memguard():
         env["GPUSHARE_MEM_LIMIT_BYTES"] = str(30 << 30)  # 8×30 < 288 GiB
         env["GPUSHARE_POD_UID"] = pod_uid
         env["T_ID"] = str(i)
-        shm_paths.append(f"/dev/shm/gpushare.memguard.{pod_uid}")
+        shm_paths.append(f"/dev/shm/gpushare.memguard.{pod_uid}.c")
         procs.append(
             subprocess.Popen(
                 [sys.executable, "-c", script],
@@ -427,6 +427,113 @@ def test_eight_tenants_shared_gpu_with_memguard():
         if os.path.exists(sp):
             os.unlink(sp)
     assert not failures, failures
+
+
+@pytest.mark.gpu
+def test_memguard_expandable_segments_vmm_budget():
+    """PyTorch's expandable_segments allocator is built on the HIP VMM
+    family (hipMemCreate/hipMemMap) — the round-1 escape hatch.  Under a
+    4 GiB budget: a 1 GiB tensor fits, a further 6 GiB tensor must OOM
+    through the interposed hipMemCreate, and usage must be repaid on
+    free."""
+    import os
+    import subprocess
+    import sys
+    import uuid as uuid_mod
+
+    import gpushare_amd
+
+    lib = os.path.join(
+        os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+    )
+    assert os.path.exists(lib), "memguard not built"
+    script = r"""
+import torch
+assert torch.cuda.is_available()
+a = torch.empty(1 << 30, dtype=torch.uint8, device="cuda:0")
+try:
+    b = torch.empty(6 << 30, dtype=torch.uint8, device="cuda:0")
+except torch.OutOfMemoryError:
+    pass
+else:
+    raise SystemExit("VMM_FAIL: 6 GiB fit inside a 4 GiB budget")
+del a
+torch.cuda.empty_cache()
+# repaid: a 3 GiB tensor fits again
+c = torch.empty(3 << 30, dtype=torch.uint8, device="cuda:0")
+c.fill_(7)
+torch.cuda.synchronize()
+print("VMM_MEMGUARD_OK")
+"""
+    pod_uid = f"vmm-{uuid_mod.uuid4().hex[:8]}"
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = lib
+    env["GPUSHARE_MEM_LIMIT_BYTES"] = str(4 << 30)
+    env["GPUSHARE_POD_UID"] = pod_uid
+    env["PYTORCH_HIP_ALLOC_CONF"] = "expandable_segments:True"
+    out = subprocess.run(
+        [sys.executable, "-c", script],
+        env=env,
+        capture_output=True,
+        text=True,
+        timeout=180,
+    )
+    shm = f"/dev/shm/gpushare.memguard.{pod_uid}.c"
+    if os.path.exists(shm):
+        os.unlink(shm)
+    assert "VMM_MEMGUARD_OK" in out.stdout, (
+        f"stdout={out.stdout!r} stderr={out.stderr[-2000:]!r}"
+    )
+
+
+@pytest.mark.gpu
+def test_memguard_per_device_cap_enforced():
+    """The per-ordinal sub-budget (multi-GPU split enforcement) must OOM
+    an allocation that fits the container total but exceeds the device's
+    cap.  On a 1-GPU box: total 6 GiB, ordinal-0 cap 2 GiB — a 3 GiB
+    tensor must fail, a 1 GiB tensor must fit."""
+    import os
+    import subprocess
+    import sys
+    import uuid as uuid_mod
+
+    import gpushare_amd
+
+    lib = os.path.join(
+        os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+    )
+    assert os.path.exists(lib), "memguard not built"
+    script = r"""
+import torch
+free, total = torch.cuda.mem_get_info()
+assert total <= (2 << 30), f"per-device clamp missing: total={total}"
+a = torch.empty(1 << 30, dtype=torch.uint8, device="cuda:0")
+try:
+    b = torch.empty(3 << 30, dtype=torch.uint8, device="cuda:0")
+except torch.OutOfMemoryError:
+    print("PER_DEVICE_CAP_OK")
+else:
+    print("PER_DEVICE_CAP_FAIL")
+"""
+    pod_uid = f"devcap-{uuid_mod.uuid4().hex[:8]}"
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = lib
+    env["GPUSHARE_MEM_LIMIT_BYTES"] = str(6 << 30)
+    env["GPUSHARE_MEM_LIMIT_BYTES_PER_DEVICE"] = str(2 << 30)
+    env["GPUSHARE_POD_UID"] = pod_uid
+    out = subprocess.run(
+        [sys.executable, "-c", script],
+        env=env,
+        capture_output=True,
+        text=True,
+        timeout=180,
+    )
+    shm = f"/dev/shm/gpushare.memguard.{pod_uid}.c"
+    if os.path.exists(shm):
+        os.unlink(shm)
+    assert "PER_DEVICE_CAP_OK" in out.stdout, (
+        f"stdout={out.stdout!r} stderr={out.stderr[-2000:]!r}"
+    )
 
 
 @pytest.mark.gpu

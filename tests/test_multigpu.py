@@ -123,6 +123,27 @@ class TestExtenderMultiGPU:
         assert sum(merged.values()) == 24
         assert set(merged) == {0, 1}
 
+    def test_assume_splits_map_across_containers(self):
+        """the map must carve the split across GPU-requesting
+        containers (spec order), not attribute everything to the first —
+        per container the units sum to its own limit, per GPU to the
+        split."""
+        kube, ext = self._extender()
+        pod = make_pod("big2c", node=NODE, mem=24, containers=[16, 8])
+        del pod["metadata"]["annotations"]
+        kube.add_pod(pod)
+        assert ext.assume(pod, NODE) == 0
+        anns = kube.get_pod("default", "big2c")["metadata"]["annotations"]
+        alloc = json.loads(anns[consts.ANN_GPUSHARE_ALLOCATION])
+        assert set(alloc) == {"c0", "c1"}
+        assert sum(alloc["c0"].values()) == 16
+        assert sum(alloc["c1"].values()) == 8
+        per_gpu: dict = {}
+        for per in alloc.values():
+            for i, u in per.items():
+                per_gpu[i] = per_gpu.get(i, 0) + u
+        assert sorted(per_gpu.values()) == [8, 16]
+
     def test_single_gpu_pod_gets_no_map(self):
         kube, ext = self._extender()
         pod = make_pod("small", node=NODE, mem=8)
@@ -197,6 +218,71 @@ class TestAllocatorMultiGPU:
         assert consts.DEV_KFD in dev_paths
         renders = [p for p in dev_paths if "renderD" in p]
         assert len(renders) == 2  # one per bound GPU
+
+    def test_multi_device_memguard_split_envs(self):
+        """Per-GPU sub-budgets: every container gets a
+        container-scoped token plus GPUSHARE_MEM_LIMIT_BYTES_PER_DEVICE
+        caps that carve the extender's split across containers — a tenant
+        can no longer concentrate the whole pod budget on one split
+        member."""
+        gpus = MockSource.from_spec("4x16GiB").devices()
+        kube = FakeKubeClient(node_name=NODE)
+        pm = PodManager(
+            kube,
+            NODE,
+            kubelet_client=kube.as_kubelet(),
+            cache_ttl=0.0,
+            kubelet_retries=0,
+            apiserver_retries=0,
+        )
+        alloc = Allocator(gpus, pm, memguard_path="/x/libmemguard.so")
+        kube.add_pod(
+            make_pod(
+                "big2c",
+                node=NODE,
+                mem=24,
+                gpu_idx=1,
+                containers=[16, 8],
+                extra_annotations={
+                    consts.ANN_GPUSHARE_ALLOCATION: json.dumps(
+                        {"c0": {"1": 16, "2": 8}}
+                    )
+                },
+            )
+        )
+        resp = alloc.allocate(_request([16, 8]))
+        c0, c1 = resp.container_responses
+        gib = 1 << 30
+        # container 0: 16 GiB total, all of it on visible ordinal 0 (GPU 1)
+        assert c0.envs[consts.ENV_MEMGUARD_CONTAINER_TOKEN] == "c0"
+        assert c0.envs[consts.ENV_MEMGUARD_LIMIT] == str(16 * gib)
+        assert c0.envs[consts.ENV_MEMGUARD_PER_DEVICE] == f"{16 * gib},0"
+        # container 1: 8 GiB total, all on visible ordinal 1 (GPU 2)
+        assert c1.envs[consts.ENV_MEMGUARD_CONTAINER_TOKEN] == "c1"
+        assert c1.envs[consts.ENV_MEMGUARD_LIMIT] == str(8 * gib)
+        assert c1.envs[consts.ENV_MEMGUARD_PER_DEVICE] == f"0,{8 * gib}"
+        # caps sum exactly to the split per GPU
+        caps0 = [int(x) for x in c0.envs[consts.ENV_MEMGUARD_PER_DEVICE].split(",")]
+        caps1 = [int(x) for x in c1.envs[consts.ENV_MEMGUARD_PER_DEVICE].split(",")]
+        assert [a + b for a, b in zip(caps0, caps1)] == [16 * gib, 8 * gib]
+
+    def test_single_gpu_memguard_has_no_per_device_env(self):
+        gpus = MockSource.from_spec("2x8GiB").devices()
+        kube = FakeKubeClient(node_name=NODE)
+        pm = PodManager(
+            kube,
+            NODE,
+            kubelet_client=kube.as_kubelet(),
+            cache_ttl=0.0,
+            kubelet_retries=0,
+            apiserver_retries=0,
+        )
+        alloc = Allocator(gpus, pm, memguard_path="/x/libmemguard.so")
+        kube.add_pod(make_pod("small", node=NODE, mem=4, gpu_idx=0))
+        resp = alloc.allocate(_request([4]))
+        c = resp.container_responses[0]
+        assert consts.ENV_MEMGUARD_PER_DEVICE not in c.envs
+        assert c.envs[consts.ENV_MEMGUARD_CONTAINER_TOKEN] == "c0"
 
     def test_gpu_split_helper(self):
         pod = make_pod(

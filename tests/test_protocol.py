@@ -197,6 +197,7 @@ class TestTLS:
         subprocess.run(
             ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
              "-days", "1", "-subj", "/CN=127.0.0.1",
+             "-addext", "subjectAltName=IP:127.0.0.1",
              "-keyout", str(key), "-out", str(cert)],
             check=True, capture_output=True,
         )
@@ -222,8 +223,72 @@ class TestTLS:
             kc = KubeletClient(address="127.0.0.1", port=srv.port, token="t")
             assert kc.get_node_running_pods() == podlist
             kc.close()
+
+            # verified TLS is the DEFAULT: the same self-signed
+            # server must be REJECTED without an explicit verify=False,
+            # and accepted when its cert is pinned as the CA
+            s_default = HttpSession(f"https://127.0.0.1:{srv.port}")
+            with pytest.raises(ssl.SSLError):
+                s_default.request("GET", "/pods/")
+            s_default.close()
+
+            s_pinned = HttpSession(
+                f"https://127.0.0.1:{srv.port}", verify=str(cert)
+            )
+            status, body = s_pinned.request("GET", "/pods/")
+            assert status == 200
+            s_pinned.close()
         finally:
             srv.stop()
+
+    def test_kubeconfig_certificate_authority_data(self, tmp_path):
+        """_auto_config must honour inline certificate-authority-data
+        (base64 PEM) — previously unparsed, which silently downgraded
+        such kubeconfigs to unverified TLS."""
+        import base64
+        import os
+        import ssl
+        import subprocess
+
+        import yaml
+
+        from gpushare_amd.cluster.kubeclient import RestKubeClient
+
+        cert = tmp_path / "ca.crt"
+        key = tmp_path / "ca.key"
+        subprocess.run(
+            ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+             "-days", "1", "-subj", "/CN=127.0.0.1",
+             "-addext", "subjectAltName=IP:127.0.0.1",
+             "-keyout", str(key), "-out", str(cert)],
+            check=True, capture_output=True,
+        )
+        kubeconfig = tmp_path / "kubeconfig"
+        kubeconfig.write_text(yaml.safe_dump({
+            "current-context": "c",
+            "contexts": [{"name": "c",
+                          "context": {"cluster": "cl", "user": "u"}}],
+            "clusters": [{"name": "cl", "cluster": {
+                "server": "https://127.0.0.1:6443",
+                "certificate-authority-data": base64.b64encode(
+                    cert.read_bytes()
+                ).decode(),
+            }}],
+            "users": [{"name": "u", "user": {"token": "tok"}}],
+        }))
+        old = os.environ.get("KUBECONFIG")
+        os.environ["KUBECONFIG"] = str(kubeconfig)
+        try:
+            server, token, verify = RestKubeClient._auto_config()
+            assert server == "https://127.0.0.1:6443"
+            assert token == "tok"
+            assert isinstance(verify, ssl.SSLContext)
+            assert verify.verify_mode == ssl.CERT_REQUIRED
+        finally:
+            if old is None:
+                os.environ.pop("KUBECONFIG", None)
+            else:
+                os.environ["KUBECONFIG"] = old
 
 
 def test_deploy_manifests_parse_and_reference_contract():
