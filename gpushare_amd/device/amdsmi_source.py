@@ -50,7 +50,12 @@ def _load_shim():
 
 
 class AmdSmiSource:
-    def __init__(self, ecc_poll_interval: float = 30.0):
+    def __init__(
+        self,
+        ecc_poll_interval: float = 30.0,
+        topology_root: str = kfd_topology.KFD_TOPOLOGY,
+        drm_root: str = kfd_topology.DRM_CLASS,
+    ):
         self._smi = _load_shim()
         if not self._smi.available():
             raise RuntimeError(
@@ -59,6 +64,8 @@ class AmdSmiSource:
             )
         self._smi.init()
         self._ecc_poll_interval = ecc_poll_interval
+        self._topology_root = topology_root
+        self._drm_root = drm_root
         self._gpus = self._enumerate()
         self._lock = threading.Lock()
 
@@ -66,7 +73,7 @@ class AmdSmiSource:
         count = self._smi.device_count()
         if count == 0:
             raise RuntimeError("amdsmi reports 0 AMD GPUs on this node")
-        topo = kfd_topology.resolve()
+        topo = kfd_topology.resolve(self._topology_root, self._drm_root)
         gpus: list[PhysicalGPU] = []
         kfd_of_index: dict[int, int] = {}
         for i in range(count):
