@@ -60,6 +60,14 @@ class PodInformer:
         self.resync_interval = resync_interval
         self.reconnect_backoff = reconnect_backoff
         self._store: dict[str, dict] = {}
+        # deletion tombstones uid -> (rv, monotonic time): a re-LIST served
+        # from a stale apiserver cache may still contain a pod whose
+        # DELETED event the watch already delivered; without a tombstone
+        # the merge would resurrect it and the allocator could match a
+        # dead pod.  Pruned by age (uids are never reused in k8s, so age
+        # only bounds memory, not correctness).
+        self._tombstones: dict[str, tuple[int, float]] = {}
+        self._tombstone_ttl = 600.0
         self._lock = threading.Lock()
         self._cv = threading.Condition(self._lock)
         self._version = 0            # bumped on every applied event / relist
@@ -146,7 +154,14 @@ class PodInformer:
                     field_selector=selector
                 ).get("items", [])
                 with self._lock:
-                    self._store = {_uid(p): p for p in snapshot}
+                    self._store = {
+                        _uid(p): p
+                        for p in snapshot
+                        if not (
+                            (tomb := self._tombstones.get(_uid(p)))
+                            and _rv(p) <= tomb[0]
+                        )
+                    }
                     self._version += 1
                     self._cv.notify_all()
                 self.relists += 1
@@ -208,9 +223,25 @@ class PodInformer:
             if etype == "DELETED":
                 if cur is None or _rv(cur) <= _rv(pod):
                     self._store.pop(uid, None)
+                    now = time.monotonic()
+                    self._tombstones[uid] = (
+                        max(_rv(pod), _rv(cur) if cur else 0),
+                        now,
+                    )
+                    if len(self._tombstones) > 10_000:
+                        cutoff = now - self._tombstone_ttl
+                        self._tombstones = {
+                            u: (rv, t)
+                            for u, (rv, t) in self._tombstones.items()
+                            if t > cutoff
+                        }
             elif etype in ("ADDED", "MODIFIED"):
-                if cur is None or _rv(cur) <= _rv(pod):
+                tomb = self._tombstones.get(uid)
+                if tomb is not None and _rv(pod) <= tomb[0]:
+                    pass  # stale replay of a pod we saw deleted
+                elif cur is None or _rv(cur) <= _rv(pod):
                     self._store[uid] = pod
+                    self._tombstones.pop(uid, None)
             elif etype == "BOOKMARK":
                 pass
             else:
@@ -226,8 +257,12 @@ class PodInformer:
         max_snap_rv = max((_rv(p) for p in snapshot), default=0)
         with self._lock:
             for uid, pod in snap.items():
+                tomb = self._tombstones.get(uid)
+                if tomb is not None and _rv(pod) <= tomb[0]:
+                    continue  # stale list still carries a deleted pod
                 if uid not in self._store or _rv(self._store[uid]) < _rv(pod):
                     self._store[uid] = pod
+                    self._tombstones.pop(uid, None)
             for uid in list(self._store):
                 if uid not in snap and _rv(self._store[uid]) <= max_snap_rv:
                     del self._store[uid]

@@ -98,6 +98,76 @@ def test_concurrent_allocates_never_double_assign():
     assert len(assigned) == 16
 
 
+def test_same_size_different_idx_env_matches_patched_pod():
+    """VERDICT r1 item 6 (SURVEY §7 hard part 1): same-size pods bound to
+    DIFFERENT GPUs, N concurrent Allocates.  The reference can mis-bind
+    here (allocate.go:78-88: env built from one candidate, patch applied
+    under the same mutex but responses paired only by size).  Our claim
+    design must guarantee, PER CALL, that the returned envs/device nodes
+    come from exactly the pod that call's ASSIGNED patch marked —
+    asymmetric annotations must never cross."""
+    for attempt in range(10):  # racy property: hammer it
+        kube = FakeKubeClient("node-a")
+        alloc = _alloc(kube)
+        # 8 same-size pods, each bound by the extender to a DIFFERENT GPU
+        for i in range(8):
+            kube.add_pod(
+                make_pod(f"p{attempt}-{i}", 4, gpu_idx=i, assume_time_ns=i)
+            )
+
+        # record which pod each call's Allocate actually patched
+        # (thread-local: mark_assigned runs on the calling thread, the
+        # result is read back in the same thread after allocate returns)
+        tls = threading.local()
+        real_mark = alloc.pods.mark_assigned
+
+        def recording_mark(pod, _real=real_mark):
+            tls.patched = pod
+            return _real(pod)
+
+        alloc.pods.mark_assigned = recording_mark
+
+        outcomes: list[tuple] = [None] * 8
+
+        def run(slot):
+            tls.patched = None
+            resp = alloc.allocate(_request([4]))
+            outcomes[slot] = (resp, tls.patched)
+
+        threads = [
+            threading.Thread(target=run, args=(s,)) for s in range(8)
+        ]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+
+        seen_idx = []
+        for resp, pod in outcomes:
+            envs = resp.container_responses[0].envs
+            idx = envs[consts.ENV_RESOURCE_INDEX]
+            assert idx != "-1", "no allocate may fail here"
+            assert pod is not None
+            pod_idx = pod["metadata"]["annotations"][
+                consts.ENV_RESOURCE_INDEX
+            ]
+            # THE property: this call's env == this call's patched pod
+            assert idx == pod_idx, (
+                f"cross-bind: env says GPU {idx} but the call patched "
+                f"pod {pod['metadata']['name']} bound to GPU {pod_idx}"
+            )
+            # device nodes must match the same GPU
+            renders = [
+                d.host_path
+                for d in resp.container_responses[0].devices
+                if "renderD" in d.host_path
+            ]
+            assert renders == [f"/dev/dri/renderD{128 + int(idx)}"]
+            seen_idx.append(idx)
+        # all 8 distinct bindings were honored exactly once
+        assert sorted(seen_idx) == [str(i) for i in range(8)]
+
+
 def test_claim_released_on_patch_failure():
     """A failed ASSIGNED patch must unclaim the pod so a retry can take it."""
     kube = FakeKubeClient("node-a")
