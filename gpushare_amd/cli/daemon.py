@@ -39,6 +39,11 @@ def parse_args(argv=None):
     p.add_argument("--deep-probe", type=float, default=0.0, metavar="SECONDS",
                    help="also run the gfx950 canary kernels (MFMA+VRAM) on "
                         "every GPU at this interval; 0 disables")
+    p.add_argument("--probe-mode", default="subprocess",
+                   choices=("subprocess", "inproc"),
+                   help="run canary probes in a short-lived child process "
+                        "(default; keeps the daemon's RSS at control-plane "
+                        "size and survives a GPU-wedging probe) or in-process")
     p.add_argument("--memory-unit", default=consts.GIB,
                    choices=list(consts.VALID_MEMORY_UNITS),
                    help="granularity of the gpu-mem resource (reference "
@@ -147,6 +152,7 @@ def main(argv=None) -> int:
             query_kubelet=args.query_kubelet,
             health_check=args.health_check,
             deep_probe_interval=args.deep_probe,
+            probe_mode=args.probe_mode,
             socket_dir=args.socket_dir,
             cache_ttl=args.cache_ttl,
             inject_devices=not args.no_inject,

@@ -18,6 +18,7 @@ from __future__ import annotations
 
 import argparse
 import sys
+import time
 
 from .. import consts
 from ..cluster import podutils
@@ -218,17 +219,32 @@ def active_pods(pods: list[dict]) -> list[dict]:
     ]
 
 
+def _with_retries(fn, retries: int = 5, interval: float = 0.1):
+    """Client-side retry budget for transient apiserver errors — the
+    kubectl plugin must not fail on first touch (reference:
+    cmd/inspect/podinfo.go:24,64-70 — 5 × 100 ms)."""
+    last: Exception = None
+    for attempt in range(retries):
+        try:
+            return fn()
+        except Exception as e:  # noqa: BLE001 — transport and API errors
+            last = e
+            if attempt < retries - 1:
+                time.sleep(interval)
+    raise last
+
+
 def build_node_infos(kube, node_name: str = "") -> list[NodeInfo]:
-    nodes = kube.list_nodes().get("items", [])
+    nodes = _with_retries(kube.list_nodes).get("items", [])
     if node_name:
         nodes = [n for n in nodes if n.get("metadata", {}).get("name") == node_name]
     nodes = [n for n in nodes if is_shared_gpu_node(n)]
     infos = []
     for node in nodes:
         name = node.get("metadata", {}).get("name", "")
-        pods = kube.list_pods(field_selector=f"spec.nodeName={name}").get(
-            "items", []
-        )
+        pods = _with_retries(
+            lambda n=name: kube.list_pods(field_selector=f"spec.nodeName={n}")
+        ).get("items", [])
         infos.append(NodeInfo(node, active_pods(pods)))
     return infos
 

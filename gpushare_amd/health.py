@@ -16,10 +16,52 @@ MI355X-native upgrades:
 
 from __future__ import annotations
 
+import json
 import logging
+import os
+import subprocess
+import sys
 import threading
 
 log = logging.getLogger(__name__)
+
+
+def probe_in_subprocess(
+    gpu_index: int, vram_probe_mb: int = 32, timeout: float = 60.0
+) -> dict:
+    """Run one canary probe in a short-lived child process.
+
+    The daemon itself must never map the HIP runtime: keeping
+    libamdhip64 (plus its device-code images) resident cost ~10× the
+    control plane's RSS in round 1 (~525 MB vs the reference's 300 Mi
+    pod, profiles/perf_evolution_gpu_box.md) — and a canary that wedges
+    the GPU now kills a disposable child, not the plugin.
+    """
+    import gpushare_amd
+
+    code = (
+        "import json, sys\n"
+        "import gpushare_amd._canary as canary\n"
+        "r = canary.probe(int(sys.argv[1]), vram_probe_mb=int(sys.argv[2]),"
+        " bandwidth=False)\n"
+        "print(json.dumps(r))\n"
+    )
+    env = dict(os.environ)
+    pkg_root = os.path.dirname(os.path.dirname(gpushare_amd.__file__))
+    env["PYTHONPATH"] = pkg_root + os.pathsep + env.get("PYTHONPATH", "")
+    proc = subprocess.run(
+        [sys.executable, "-c", code, str(gpu_index), str(vram_probe_mb)],
+        capture_output=True,
+        text=True,
+        timeout=timeout,
+        env=env,
+    )
+    if proc.returncode != 0:
+        raise RuntimeError(
+            f"canary subprocess rc={proc.returncode}: "
+            f"{proc.stderr.strip()[-500:]}"
+        )
+    return json.loads(proc.stdout.strip().splitlines()[-1])
 
 
 class HealthMonitor:
@@ -30,12 +72,14 @@ class HealthMonitor:
         deep_probe_interval: float = 0.0,   # 0 = passive only
         probe_vram_mb: int = 32,
         event_recorder=None,
+        probe_mode: str = "subprocess",     # "subprocess" | "inproc"
     ):
         self.source = source
         self.plugin = plugin
         self.deep_probe_interval = deep_probe_interval
         self.probe_vram_mb = probe_vram_mb
         self.events = event_recorder
+        self.probe_mode = probe_mode
         self._stop = threading.Event()
         self._threads: list[threading.Thread] = []
         self._probe_failed: set[int] = set()
@@ -77,12 +121,22 @@ class HealthMonitor:
         except Exception as e:  # noqa: BLE001
             log.error("passive health watcher died: %s", e)
 
-    def _probe_loop(self) -> None:
-        try:
+    def _probe_once(self, gpu_index: int) -> dict:
+        if self.probe_mode == "inproc":
             import gpushare_amd._canary as canary
-        except ImportError as e:
-            log.error("deep probe requested but _canary not built: %s", e)
-            return
+
+            return canary.probe(
+                gpu_index, vram_probe_mb=self.probe_vram_mb, bandwidth=False
+            )
+        return probe_in_subprocess(gpu_index, self.probe_vram_mb)
+
+    def _probe_loop(self) -> None:
+        if self.probe_mode == "inproc":
+            try:
+                import gpushare_amd._canary  # noqa: F401
+            except ImportError as e:
+                log.error("deep probe requested but _canary not built: %s", e)
+                return
         from . import metrics
 
         while not self._stop.wait(self.deep_probe_interval):
@@ -95,12 +149,13 @@ class HealthMonitor:
                 if self._stop.is_set():
                     return
                 try:
-                    result = canary.probe(
-                        gpu.index, vram_probe_mb=self.probe_vram_mb,
-                        bandwidth=False,
-                    )
+                    result = self._probe_once(gpu.index)
                     ok = bool(result.get("ok"))
-                except RuntimeError as e:
+                except (
+                    RuntimeError,
+                    subprocess.TimeoutExpired,
+                    ValueError,
+                ) as e:
                     log.error("canary probe GPU %d errored: %s", gpu.index, e)
                     ok = False
                     result = {"error": str(e)}

@@ -52,6 +52,9 @@ class ManagerOptions:
                                  # "" disables VRAM budget enforcement
     allow_oversize_inventory: bool = False
     trace_file: str = ""         # JSONL per-Allocate trace (debugging)
+    probe_mode: str = "subprocess"  # canary probes in a short-lived child
+                                    # (daemon never maps the HIP runtime);
+                                    # "inproc" keeps the r1 behavior
 
 
 class SharedGPUManager:
@@ -125,6 +128,7 @@ class SharedGPUManager:
                 self.plugin,
                 deep_probe_interval=self.opt.deep_probe_interval,
                 event_recorder=EventRecorder(self.kube, self.node_name),
+                probe_mode=self.opt.probe_mode,
             )
             self.health.start()
 

@@ -209,3 +209,45 @@ def test_gpushare_top_json_output():
     assert top.main(["-o", "json"], source=Src(), out=out) == 0
     d = json.loads(out.getvalue())
     assert d["gpus"][0]["vram_used_bytes"] == 1 << 30
+
+
+def test_inspect_retries_transient_apiserver_errors(capsys):
+    """Reference parity (podinfo.go:64-70): a transient error on the
+    first list must be retried, not surfaced to the kubectl user."""
+    from gpushare_amd.cli import inspect as inspect_cli
+
+    kube = FakeKubeClient("node-a")
+    kube.patch_node_status(
+        "node-a",
+        {"status": {"allocatable": {consts.RESOURCE_NAME: "288"},
+                    "capacity": {consts.RESOURCE_NAME: "288"}}},
+    )
+    fails = {"n": 2}
+    real_list_nodes = kube.list_nodes
+
+    def flaky_list_nodes():
+        if fails["n"] > 0:
+            fails["n"] -= 1
+            raise ConnectionError("transient apiserver hiccup")
+        return real_list_nodes()
+
+    kube.list_nodes = flaky_list_nodes
+    infos = inspect_cli.build_node_infos(kube)
+    assert [i.name for i in infos] == ["node-a"]
+    assert fails["n"] == 0
+
+
+def test_inspect_retry_budget_exhausts():
+    from gpushare_amd.cli import inspect as inspect_cli
+
+    calls = {"n": 0}
+
+    def always_fails():
+        calls["n"] += 1
+        raise ConnectionError("down")
+
+    import pytest as _pytest
+
+    with _pytest.raises(ConnectionError):
+        inspect_cli._with_retries(always_fails, retries=5, interval=0.001)
+    assert calls["n"] == 5

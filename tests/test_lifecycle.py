@@ -222,3 +222,74 @@ def test_allocation_state_survives_plugin_restart(tmp_socket_dir):
         mgr.shutdown()
         t.join(timeout=5)
         kubelet.stop()
+
+
+def test_health_monitor_subprocess_probe_flip_and_recover(
+    tmp_socket_dir, monkeypatch
+):
+    """Deep probe in subprocess mode (r2 default: the daemon never maps
+    the HIP runtime): a failing probe flips the GPU Unhealthy, a passing
+    one recovers it — without gpushare_amd._canary ever being imported
+    into this process."""
+    import sys
+
+    from gpushare_amd import health as health_mod
+    from gpushare_amd.allocator import Allocator
+    from gpushare_amd.cluster.podmanager import PodManager
+    from gpushare_amd.deviceplugin.server import GPUSharePlugin
+
+    kube = FakeKubeClient("node-a")
+    source = MockSource.from_spec("2x8GiB")
+    gpus = source.devices()
+    pm = PodManager(
+        kube, "node-a", kubelet_client=kube.as_kubelet(), cache_ttl=0
+    )
+    plugin = GPUSharePlugin(
+        gpus, Allocator(gpus, pm), socket_dir=tmp_socket_dir
+    )
+
+    state = {"fail": True}
+
+    def fake_probe(gpu_index, vram_probe_mb=32, timeout=60.0):
+        if state["fail"] and gpu_index == 1:
+            raise RuntimeError("canary subprocess rc=1: hipErrorNoDevice")
+        return {"ok": True, "mfma_ok": True}
+
+    monkeypatch.setattr(health_mod, "probe_in_subprocess", fake_probe)
+    mon = HealthMonitor(
+        source, plugin, deep_probe_interval=0.02, probe_mode="subprocess"
+    )
+    mon.start()
+    try:
+        deadline = time.monotonic() + 5
+        while plugin._unhealthy_gpus != {1} and time.monotonic() < deadline:
+            time.sleep(0.01)
+        assert plugin._unhealthy_gpus == {1}
+
+        state["fail"] = False
+        while plugin._unhealthy_gpus and time.monotonic() < deadline:
+            time.sleep(0.01)
+        assert plugin._unhealthy_gpus == set()
+        # the probe never pulled the HIP runtime into THIS process
+        assert "gpushare_amd._canary" not in sys.modules
+    finally:
+        mon.stop()
+
+
+def test_probe_in_subprocess_fails_loudly_without_gpu():
+    """On a GPU-less host the child must exit nonzero and surface its
+    stderr — never a silent healthy verdict."""
+    import pytest as _pytest
+
+    from gpushare_amd import health as health_mod
+
+    try:
+        import gpushare_amd._canary  # noqa: F401
+    except ImportError:
+        _pytest.skip("_canary extension not built")
+    import torch
+
+    if torch.cuda.is_available():
+        _pytest.skip("GPU present; covered by the gpu-marked test")
+    with _pytest.raises(RuntimeError, match="canary subprocess"):
+        health_mod.probe_in_subprocess(0, vram_probe_mb=1)
