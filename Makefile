@@ -24,6 +24,7 @@ scale: build                # co-location + inventory-scale experiments
 
 lint:
 	$(PY) -m compileall -q gpushare_amd tests bench.py __graft_entry__.py
+	$(PY) tools/lint.py
 
 image:
 	docker build -t gpushare/amd-device-plugin:latest .

@@ -84,8 +84,88 @@ def parse_args(argv=None):
                         "outcome, per-stage ms) for debugging")
     p.add_argument("--metrics-port", type=int, default=0,
                    help="serve Prometheus /metrics on this port (0 = off)")
+    p.add_argument("--selftest", action="store_true",
+                   help="one full register→ListAndWatch→Allocate pass "
+                        "against an in-process stub kubelet with mock "
+                        "GPUs, then exit 0/1 (container HEALTHCHECK / CI "
+                        "smoke; no cluster, no GPU needed)")
     p.add_argument("-v", "--verbose", action="count", default=0)
     return p.parse_args(argv)
+
+
+def selftest() -> int:
+    """Image smoke: the whole plumbing path on fake devices.
+
+    Exercises exactly BASELINE config 1 (mock GPU + stub kubelet socket):
+    gRPC registration, the pre-encoded ListAndWatch inventory, and one
+    Allocate with env + device-node injection.  Returns 0 on success.
+    """
+    import tempfile
+
+    from ..allocator import Allocator
+    from ..cluster.kubeclient import FakeKubeClient
+    from ..cluster.podmanager import PodManager
+    from ..device.mock_source import MockSource
+    from ..deviceplugin.server import GPUSharePlugin
+    from ..deviceplugin.stubkubelet import StubKubelet
+
+    log = logging.getLogger("selftest")
+    with tempfile.TemporaryDirectory(prefix="gpushare-st-") as socket_dir:
+        kube = FakeKubeClient(node_name="selftest-node")
+        pm = PodManager(
+            kube, "selftest-node", kubelet_client=kube.as_kubelet(),
+            cache_ttl=0.0, kubelet_retries=0, kubelet_retry_interval=0.0,
+            apiserver_retries=0, apiserver_retry_interval=0.0,
+        )
+        gpus = MockSource.from_spec("1x8GiB").devices()
+        plugin = GPUSharePlugin(gpus, Allocator(gpus, pm),
+                                socket_dir=socket_dir)
+        kubelet = StubKubelet(socket_dir)
+        kubelet.start()
+        try:
+            plugin.serve()
+            client = kubelet.wait_for_plugin(consts.RESOURCE_NAME)
+            devices = client.wait_for_devices(min_count=8)
+            if len(devices) != 8:
+                log.error("selftest: expected 8 fake devices, got %d",
+                          len(devices))
+                return 1
+            # hand-rolled assumed pod (no test helpers in the image)
+            import time as _time
+            kube.add_pod({
+                "metadata": {
+                    "name": "selftest-pod", "namespace": "default",
+                    "uid": "selftest-uid",
+                    "annotations": {
+                        consts.ENV_RESOURCE_INDEX: "0",
+                        consts.ENV_RESOURCE_ASSUME_TIME: str(_time.time_ns()),
+                        consts.ENV_ASSIGNED_FLAG: "false",
+                    },
+                },
+                "spec": {
+                    "nodeName": "selftest-node",
+                    "containers": [{
+                        "name": "c0",
+                        "resources": {"limits": {consts.RESOURCE_NAME: "4"}},
+                    }],
+                },
+                "status": {"phase": "Pending"},
+            })
+            ids = sorted(client.wait_for_devices(8))[:4]
+            resp = client.allocate([ids])
+            envs = resp.container_responses[0].envs
+            if envs[consts.ENV_RESOURCE_INDEX] != "0":
+                log.error("selftest: bad allocate envs: %s", dict(envs))
+                return 1
+            log.info("selftest OK: registered, streamed 8 devices, "
+                     "allocated 4 units on GPU 0")
+            return 0
+        except Exception as e:  # noqa: BLE001
+            log.error("selftest failed: %s", e)
+            return 1
+        finally:
+            plugin.stop()
+            kubelet.stop()
 
 
 def main(argv=None) -> int:
@@ -96,6 +176,9 @@ def main(argv=None) -> int:
         stream=sys.stderr,
     )
     log = logging.getLogger("daemon")
+
+    if args.selftest:
+        return selftest()
 
     node_name = os.environ.get("NODE_NAME")
     if not node_name:

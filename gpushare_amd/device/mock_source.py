@@ -17,7 +17,7 @@ from __future__ import annotations
 
 import queue
 import re
-from typing import Iterable, Optional
+from typing import Iterable
 
 from . import HealthEvent, PhysicalGPU
 

@@ -137,7 +137,6 @@ def test_memguard_lib_loads_and_reads_env(tmp_path):
     """CPU-side smoke of libgpushare_memguard.so: the lib must dlopen
     anywhere (links only libdl/libc), parse the env budget, and map its
     pod-scoped slot table (no HIP runtime involved)."""
-    import ctypes
     import os
     import subprocess
     import sys
