@@ -257,3 +257,69 @@ def test_memguard_per_device_limits_parsed():
     if os.path.exists(shm):
         os.unlink(shm)
     assert "PER_DEVICE_OK" in out.stdout, out.stderr
+
+
+def test_memguard_fork_child_gets_own_slot():
+    """After fork() each process must own its own slot (a child never
+    charges or repays its parent's reservation): two live pids appear in
+    the container's table.  Slot layout (memguard.cpp struct Slot):
+    pid(i32) pad(i32) used(i64) used_dev[8](i64) = 80 bytes."""
+    import os
+    import struct
+    import subprocess
+    import sys
+
+    import gpushare_amd
+
+    lib = os.path.join(
+        os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+    )
+    if not os.path.exists(lib):
+        import pytest
+
+        pytest.skip("memguard not built")
+    uid = f"forktest-{os.getpid()}"
+    code = f"""
+import ctypes, os, sys, time
+l = ctypes.CDLL({lib!r})
+l.gpushare_memguard_used.restype = ctypes.c_int64
+assert l.gpushare_memguard_used() == 0      # parent binds its slot
+pid = os.fork()
+if pid == 0:
+    assert l.gpushare_memguard_used() == 0  # child binds a NEW slot
+    time.sleep(30)                          # hold the slot until killed
+    os._exit(0)
+time.sleep(0.5)                             # let the child bind
+print("PARENT", os.getpid(), "CHILD", pid, flush=True)
+sys.stdout.flush()
+time.sleep(5)
+os.kill(pid, 9)
+"""
+    env = dict(os.environ)
+    env["GPUSHARE_MEM_LIMIT_BYTES"] = str(1 << 30)
+    env["GPUSHARE_POD_UID"] = uid
+    shm = f"/dev/shm/gpushare.memguard.{uid}.c"
+    proc = subprocess.Popen(
+        [sys.executable, "-c", code], env=env,
+        stdout=subprocess.PIPE, text=True,
+    )
+    try:
+        line = proc.stdout.readline()
+        assert line.startswith("PARENT"), line
+        _, ppid, _, cpid = line.split()
+        with open(shm, "rb") as fh:
+            table = fh.read()
+        slot_size = 80
+        pids = set()
+        for off in range(0, len(table), slot_size):
+            (pid,) = struct.unpack_from("<i", table, off)
+            if pid:
+                pids.add(pid)
+        assert pids == {int(ppid), int(cpid)}, (
+            f"expected parent+child slots, got {pids}"
+        )
+    finally:
+        proc.kill()
+        proc.wait(timeout=10)
+        if os.path.exists(shm):
+            os.unlink(shm)
