@@ -555,3 +555,99 @@ def test_golden_client_full_flow(golden_harness):
         calls["prestart"](psr, timeout=5)
     finally:
         channel.close()
+
+
+# --------------------------------------------------------------------------- #
+# 6. property-based cross-serialization (hypothesis)
+# --------------------------------------------------------------------------- #
+try:
+    from hypothesis import given, settings, strategies as st
+
+    _HAVE_HYPOTHESIS = True
+except ImportError:  # pragma: no cover
+    _HAVE_HYPOTHESIS = False
+
+if _HAVE_HYPOTHESIS:
+    _ids = st.text(
+        alphabet=st.characters(
+            whitelist_categories=("Lu", "Ll", "Nd"), max_codepoint=127
+        ),
+        max_size=63,
+    )
+    _envs = st.dictionaries(_ids.filter(bool), _ids, max_size=8)
+    _mounts = st.lists(
+        st.fixed_dictionaries(
+            {
+                "container_path": _ids,
+                "host_path": _ids,
+                "read_only": st.booleans(),
+            }
+        ),
+        max_size=4,
+    )
+    _devspecs = st.lists(
+        st.fixed_dictionaries(
+            {
+                "container_path": _ids,
+                "host_path": _ids,
+                "permissions": st.sampled_from(["r", "rw", "mrw", ""]),
+            }
+        ),
+        max_size=4,
+    )
+
+    @settings(max_examples=200, deadline=None)
+    @given(
+        devices=st.lists(
+            st.fixed_dictionaries(
+                {
+                    "ID": _ids,
+                    "health": st.sampled_from(["Healthy", "Unhealthy", ""]),
+                }
+            ),
+            max_size=20,
+        )
+    )
+    def test_prop_list_and_watch_bytes_match(golden_module_classes, devices):
+        classes = golden_module_classes
+        mine = _fill(
+            ours.ListAndWatchResponse(), {"devices": devices}
+        ).SerializeToString(deterministic=True)
+        theirs = _fill(
+            classes["ListAndWatchResponse"](), {"devices": devices}
+        ).SerializeToString(deterministic=True)
+        assert mine == theirs
+
+    @settings(max_examples=200, deadline=None)
+    @given(envs=_envs, mounts=_mounts, devspecs=_devspecs, anns=_envs)
+    def test_prop_allocate_response_bytes_match(
+        golden_module_classes, envs, mounts, devspecs, anns
+    ):
+        classes = golden_module_classes
+        values = {
+            "container_responses": [
+                {
+                    "envs": envs,
+                    "mounts": mounts,
+                    "devices": devspecs,
+                    "annotations": anns,
+                }
+            ]
+        }
+        mine = _fill(ours.AllocateResponse(), values).SerializeToString(
+            deterministic=True
+        )
+        theirs = _fill(
+            classes["AllocateResponse"](), values
+        ).SerializeToString(deterministic=True)
+        assert mine == theirs
+        # cross-parse: protoc-derived class decodes our bytes losslessly
+        parsed = classes["AllocateResponse"]()
+        parsed.ParseFromString(mine)
+        assert parsed.SerializeToString(deterministic=True) == mine
+
+
+@pytest.fixture(scope="module")
+def golden_module_classes(golden):
+    _, classes = golden
+    return classes
