@@ -13,6 +13,7 @@ walker must exceed because ROCm minors really are non-sequential.
 from __future__ import annotations
 
 import json
+import os
 
 import pytest
 
@@ -193,3 +194,72 @@ class TestExtenderOnFixture:
         assert len(split) == 4
         assert sum(split.values()) == 1000
         assert state._xgmi_edges(tuple(split)) == 6  # complete K4
+
+
+# --------------------------------------------------------------------------- #
+# REAL captured topology (tests/fixtures/kfd_snapshot_mi355x_1gpu)
+# --------------------------------------------------------------------------- #
+REAL_SNAPSHOT = os.path.join(
+    os.path.dirname(__file__), "fixtures", "kfd_snapshot_mi355x_1gpu"
+)
+
+
+class TestRealSnapshot:
+    """Ground truth from a gpurun MI355X box: 1 visible GPU out of 8,
+    masked peers with EMPTY gpu_id files, render minor 144, 287.98 GiB."""
+
+    def _resolve(self):
+        return kfd_topology.resolve(
+            os.path.join(REAL_SNAPSHOT, "kfd", "topology", "nodes"),
+            os.path.join(REAL_SNAPSHOT, "drm"),
+        )
+
+    def test_exactly_one_visible_gpu(self):
+        topo = self._resolve()
+        assert list(topo) == [23660]
+
+    def test_masked_peers_not_gpus(self):
+        """Empty gpu_id files (the 7 masked OAMs) must parse as CPU-side
+        nodes, never as 0-byte GPUs."""
+        nodes = kfd_topology.read_topology(
+            os.path.join(REAL_SNAPSHOT, "kfd", "topology", "nodes")
+        )
+        gpus = [n for n in nodes if n.is_gpu]
+        assert len(gpus) == 1
+        assert gpus[0].gpu_id == 23660
+
+    def test_real_gpu_properties(self):
+        t = self._resolve()[23660]
+        assert t.render_path == "/dev/dri/renderD144"  # NOT 128+index
+        assert t.gfx_target_version == 90500
+        assert t.vram_bytes == 309_220_868_096  # 287.98 GiB, not clean 288
+        # xGMI links point at MASKED nodes -> no resolvable peers
+        assert t.xgmi_peer_gpu_ids == []
+
+    def test_grain_count_287(self):
+        """Real HBM size floors to 287 whole GiB grains (what round-1 GPU
+        runs advertised)."""
+        from gpushare_amd.device import PhysicalGPU
+        from gpushare_amd.device.fakedev import FakeDeviceTable
+
+        t = self._resolve()[23660]
+        gpu = PhysicalGPU(
+            index=0, uuid="amd-real", memory_bytes=t.vram_bytes
+        )
+        table = FakeDeviceTable.build([gpu], consts.GIB)
+        assert len(table) == 287
+        assert all(len(i) <= 63 for i in table.ids)
+
+    def test_masked_mesh_visible_in_raw_links(self):
+        """The full 7-link xGMI mesh is present in the raw io_links even
+        when peers are masked — the basis for the 8-GPU synthetic fixture's
+        shape."""
+        nodes = {
+            n.node_id: n
+            for n in kfd_topology.read_topology(
+                os.path.join(REAL_SNAPSHOT, "kfd", "topology", "nodes")
+            )
+        }
+        gpu = nodes[4]
+        assert sorted(gpu.xgmi_peer_nodes) == [2, 3, 5, 6, 7, 8, 9]
+        assert gpu.pcie_peer_nodes == [0]
