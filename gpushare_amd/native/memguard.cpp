@@ -234,19 +234,7 @@ int64_t total_used(bool sweep_dead, int64_t* dev_sum = nullptr) {
     return sum;
 }
 
-int current_device() {
-    static GetDeviceFn fn = nullptr;
-    if (fn == nullptr) {
-        void* sym = dlsym(RTLD_NEXT, "hipGetDevice");
-        if (sym == nullptr) sym = dlsym(RTLD_DEFAULT, "hipGetDevice");
-        fn = reinterpret_cast<GetDeviceFn>(sym);
-    }
-    int dev = 0;
-    if (fn != nullptr && fn(&dev) == HIP_SUCCESS && dev >= 0 &&
-        dev < MAX_DEV)
-        return dev;
-    return 0;
-}
+int current_device();  // fwd: defined after real<> (needs hip_handle)
 
 bool over_dev_budget(int dev, const int64_t* dev_sum) {
     return dev < g_ndev_limits && g_dev_limit[dev] >= 0 &&
@@ -334,6 +322,18 @@ Fn real(const char* name) {
     if (sym == nullptr && hip_handle() != nullptr)
         sym = dlsym(hip_handle(), name);
     return reinterpret_cast<Fn>(sym);
+}
+
+int current_device() {
+    // same resolution path as the interposed entry points: PyTorch loads
+    // libamdhip64 RTLD_LOCAL, so RTLD_NEXT alone never sees hipGetDevice
+    // and per-device charging would silently pin to ordinal 0
+    static GetDeviceFn fn = real<GetDeviceFn>("hipGetDevice");
+    int dev = 0;
+    if (fn != nullptr && fn(&dev) == HIP_SUCCESS && dev >= 0 &&
+        dev < MAX_DEV)
+        return dev;
+    return 0;
 }
 
 int guarded_alloc(MallocFn fn, void** ptr, size_t size) {
