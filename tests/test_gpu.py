@@ -165,6 +165,100 @@ def test_e2e_allocate_and_tenant_workloads(source, tmp_path):
         kubelet.stop()
 
 
+def test_allocate_envs_drive_memguard_end_to_end(source, tmp_path):
+    """The FULL enforcement chain on a real MI355X: a pod's Allocate
+    response (not hand-crafted envs) configures the tenant — LD_PRELOAD
+    mount + budget + pod/container scoping all come from the plugin, and
+    the budget must then actually bind a PyTorch tenant."""
+    from gpushare_amd.allocator import Allocator
+    from gpushare_amd.cluster.kubeclient import FakeKubeClient
+    from gpushare_amd.cluster.podmanager import PodManager
+    from gpushare_amd.deviceplugin.server import GPUSharePlugin
+    from gpushare_amd.deviceplugin.stubkubelet import StubKubelet
+    from helpers import make_pod
+
+    import gpushare_amd
+
+    lib = os.path.join(
+        os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+    )
+    assert os.path.exists(lib), "memguard not built"
+
+    gpus = source.devices()
+    sockdir = str(tmp_path / "dp")
+    os.makedirs(sockdir)
+    kube = FakeKubeClient(node_name="gpu-node")
+    pm = PodManager(
+        kube, "gpu-node", kubelet_client=kube.as_kubelet(), cache_ttl=0.0
+    )
+    plugin = GPUSharePlugin(
+        gpus, Allocator(gpus, pm, memguard_path=lib), socket_dir=sockdir
+    )
+    kubelet = StubKubelet(sockdir)
+    kubelet.start()
+    shm = None
+    try:
+        plugin.serve()
+        client = kubelet.wait_for_plugin(consts.RESOURCE_NAME)
+        all_ids = sorted(client.wait_for_devices(min_count=30))
+
+        kube.add_pod(make_pod("budget-pod", 30, gpu_idx=0, node="gpu-node"))
+        resp = client.allocate([all_ids[:30]])
+        c = resp.container_responses[0]
+        envs = dict(c.envs)
+        # the response's enforcement contract
+        assert envs[consts.ENV_MEMGUARD_LIMIT] == str(30 << 30)
+        assert envs["LD_PRELOAD"] == consts.MEMGUARD_CONTAINER_PATH
+        mounts = [(m.host_path, m.container_path) for m in c.mounts]
+        assert (lib, consts.MEMGUARD_CONTAINER_PATH) in mounts
+        pod_uid = envs[consts.ENV_MEMGUARD_POD_UID]
+        token = envs[consts.ENV_MEMGUARD_CONTAINER_TOKEN]
+        shm = f"/dev/shm/gpushare.memguard.{pod_uid}.{token}"
+
+        script = (
+            "import torch;"
+            "free, total = torch.cuda.mem_get_info();"
+            "assert total <= 30<<30, f'not clamped: {total}';"
+            "a = torch.empty(8<<30, dtype=torch.uint8, device='cuda:0');"
+            "exc = None\n"
+            "try:\n"
+            "    b = torch.empty(31<<30, dtype=torch.uint8, device='cuda:0')\n"
+            "except torch.OutOfMemoryError:\n"
+            "    print('E2E_MEMGUARD_OK', flush=True)\n"
+            "else:\n"
+            "    print('E2E_MEMGUARD_FAIL', flush=True)\n"
+        )
+        env = dict(os.environ)
+        for key in (
+            consts.ENV_MEMGUARD_LIMIT,
+            consts.ENV_MEMGUARD_POD_UID,
+            consts.ENV_MEMGUARD_CONTAINER_TOKEN,
+            consts.ENV_ROCR_VISIBLE,
+            consts.ENV_HIP_VISIBLE,
+        ):
+            if key in envs:
+                env[key] = envs[key]
+        # in k8s the kubelet bind-mounts host_path at container_path; here
+        # both are this host, so preload the mount's SOURCE directly
+        env["LD_PRELOAD"] = lib
+        out = subprocess.run(
+            [sys.executable, "-c", script],
+            env=env,
+            capture_output=True,
+            text=True,
+            timeout=240,
+            cwd=REPO,
+        )
+        assert "E2E_MEMGUARD_OK" in out.stdout, (
+            f"stdout={out.stdout!r} stderr={out.stderr[-1500:]!r}"
+        )
+    finally:
+        if shm and os.path.exists(shm):
+            os.unlink(shm)
+        plugin.stop()
+        kubelet.stop()
+
+
 def test_manager_e2e_on_gpu(source, tmp_path):
     """Full lifecycle manager with the real device source on an MI355X."""
     import threading
