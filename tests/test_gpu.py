@@ -170,6 +170,7 @@ def test_allocate_envs_drive_memguard_end_to_end(source, tmp_path):
     response (not hand-crafted envs) configures the tenant — LD_PRELOAD
     mount + budget + pod/container scoping all come from the plugin, and
     the budget must then actually bind a PyTorch tenant."""
+    from gpushare_amd import consts
     from gpushare_amd.allocator import Allocator
     from gpushare_amd.cluster.kubeclient import FakeKubeClient
     from gpushare_amd.cluster.podmanager import PodManager
