@@ -84,6 +84,11 @@ def parse_args(argv=None):
                         "outcome, per-stage ms) for debugging")
     p.add_argument("--metrics-port", type=int, default=0,
                    help="serve Prometheus /metrics on this port (0 = off)")
+    p.add_argument("--numa-topology", action="store_true",
+                   help="advertise each grain's NUMA domain via the modern "
+                        "Device.topology field (kubelet TopologyManager "
+                        "NUMA alignment; requires k8s >= 1.17 semantics — "
+                        "older kubelets skip the field)")
     p.add_argument("--selftest", action="store_true",
                    help="one full register→ListAndWatch→Allocate pass "
                         "against an in-process stub kubelet with mock "
@@ -243,6 +248,7 @@ def main(argv=None) -> int:
             memguard_path=memguard_path,
             allow_oversize_inventory=args.allow_oversize_inventory,
             trace_file=args.trace_file,
+            numa_topology=args.numa_topology,
         ),
     )
     mgr.install_signal_handlers()

@@ -44,6 +44,7 @@ class FakeDeviceTable:
     ranges: dict[int, tuple[int, int]]  # GPU index -> [start, end) into ids
     unit: str
     units_per_gpu: list[int]
+    numa_of: list[int]             # fake idx -> NUMA node (-1 unknown)
 
     @classmethod
     def build(cls, gpus: list[PhysicalGPU], unit: str) -> "FakeDeviceTable":
@@ -53,15 +54,18 @@ class FakeDeviceTable:
         gpu_of: dict[str, int] = {}
         ranges: dict[int, tuple[int, int]] = {}
         units_per_gpu: list[int] = []
+        numa_of: list[int] = []
         for g in gpus:
             n = g.mem_units(unit)
             if n <= 0:
                 raise ValueError(f"GPU {g.index} reports no memory")
             start = len(ids)
+            numa = getattr(g, "numa_node", -1)
             for j in range(n):
                 fid = fake_id(g.uuid, j)
                 ids.append(fid)
                 gpu_of[fid] = g.index
+                numa_of.append(numa if numa is not None else -1)
             # keyed by the PLUGIN index, not list position — robust to
             # non-contiguous index sets (e.g. partitioned/filtered nodes)
             ranges[g.index] = (start, len(ids))
@@ -74,6 +78,7 @@ class FakeDeviceTable:
             ranges=ranges,
             unit=unit,
             units_per_gpu=units_per_gpu,
+            numa_of=numa_of,
         )
 
     def __len__(self) -> int:
@@ -87,9 +92,13 @@ class FakeDeviceTable:
         return range(start, end)
 
 
-def encode_list_python(ids: list[str], unhealthy: set[int]) -> bytes:
+def encode_list_python(
+    ids: list[str], unhealthy: set[int], numa: list[int] | None = None
+) -> bytes:
     """Pure-python ListAndWatchResponse encoder — fallback and test oracle
-    for the native codec; byte-identical output."""
+    for the native codec; byte-identical output.  ``numa`` (optional,
+    per fake device, -1 = omit) adds Device.topology NUMA hints — the
+    modern kubelet TopologyManager field (upstream api.proto field 3)."""
 
     def varint(v: int) -> bytes:
         out = bytearray()
@@ -104,26 +113,38 @@ def encode_list_python(ids: list[str], unhealthy: set[int]) -> bytes:
         health = b"Unhealthy" if k in unhealthy else b"Healthy"
         idb = fid.encode()
         dev = b"\x0a" + varint(len(idb)) + idb + b"\x12" + varint(len(health)) + health
+        if numa is not None and numa[k] >= 0:
+            # NUMANode{ID=<n>} wrapped in TopologyInfo.nodes wrapped in
+            # Device.topology (field 3, wire type 2)
+            numanode = b"\x08" + varint(numa[k])
+            topo = b"\x0a" + varint(len(numanode)) + numanode
+            dev += b"\x1a" + varint(len(topo)) + topo
         out += b"\x0a" + varint(len(dev)) + dev
     return bytes(out)
 
 
-def make_codec(ids: list[str]):
-    """Native codec if built, else a python shim with the same interface."""
+def make_codec(ids: list[str], numa: list[int] | None = None):
+    """Native codec if built, else a python shim with the same interface.
+    ``numa``: optional per-device NUMA node (-1 = omit topology)."""
     try:
         from .. import _devlist
 
+        if numa is not None:
+            return _devlist.DeviceListCodec(ids, numa)
         return _devlist.DeviceListCodec(ids)
     except ImportError:
 
         class _PyCodec:
-            def __init__(self, ids_):
+            def __init__(self, ids_, numa_):
                 self._ids = list(ids_)
+                self._numa = list(numa_) if numa_ is not None else None
 
             def __len__(self):
                 return len(self._ids)
 
             def encode(self, unhealthy=()):
-                return encode_list_python(self._ids, set(unhealthy))
+                return encode_list_python(
+                    self._ids, set(unhealthy), self._numa
+                )
 
-        return _PyCodec(ids)
+        return _PyCodec(ids, numa)

@@ -43,6 +43,7 @@ class GPUSharePlugin:
         socket_name: str = consts.SERVER_SOCK_NAME,
         max_workers: int = 64,  # each ListAndWatch stream pins a worker thread
         allow_oversize_inventory: bool = False,
+        numa_topology: bool = False,
     ):
         self.gpus = gpus
         self.allocator = allocator
@@ -54,7 +55,16 @@ class GPUSharePlugin:
         self.max_workers = max_workers
 
         self.table = FakeDeviceTable.build(gpus, unit)
-        self._codec = make_codec(self.table.ids)
+        # --numa-topology: advertise each grain's NUMA domain via the
+        # modern Device.topology field so the kubelet TopologyManager can
+        # NUMA-align gpu-mem with cpu/memory (matters on a 2-socket 8-OAM
+        # MI355X box: 4 GPUs per socket).  Off by default — pre-1.17
+        # kubelets skip the unknown field, but byte-parity with the
+        # reference's wire traffic is the default stance.
+        self._codec = make_codec(
+            self.table.ids,
+            numa=self.table.numa_of if numa_topology else None,
+        )
         metrics.observe_inventory(len(self.table))
         # a payload over the kubelet's 4 MiB gRPC receive default fails
         # RESOURCE_EXHAUSTED on the kubelet side — at MiB grain one 288 GiB

@@ -19,6 +19,7 @@ _LABEL_OPTIONAL = descriptor_pb2.FieldDescriptorProto.LABEL_OPTIONAL
 _LABEL_REPEATED = descriptor_pb2.FieldDescriptorProto.LABEL_REPEATED
 _TYPE_STRING = descriptor_pb2.FieldDescriptorProto.TYPE_STRING
 _TYPE_BOOL = descriptor_pb2.FieldDescriptorProto.TYPE_BOOL
+_TYPE_INT64 = descriptor_pb2.FieldDescriptorProto.TYPE_INT64
 _TYPE_MESSAGE = descriptor_pb2.FieldDescriptorProto.TYPE_MESSAGE
 
 
@@ -79,10 +80,26 @@ def _build_file() -> descriptor_pb2.FileDescriptorProto:
                     "devices", 1, _TYPE_MESSAGE, _LABEL_REPEATED, t("Device")
                 ),
             ),
+            # Device.topology (field 3) + TopologyInfo/NUMANode are the
+            # MODERN upstream additions (k8s ≥1.17 api.proto; the vendored
+            # snapshot predates them): NUMA hints for the kubelet
+            # TopologyManager.  Field numbers match upstream exactly; a
+            # pre-1.17 kubelet skips field 3 as an unknown (proto3), so
+            # sending it is always safe — the plugin still gates it
+            # behind --numa-topology for byte-parity-by-default.
             _msg(
                 "Device",
                 _field("ID", 1, _TYPE_STRING),
                 _field("health", 2, _TYPE_STRING),
+                _field("topology", 3, _TYPE_MESSAGE, type_name=t("TopologyInfo")),
+            ),
+            _msg(
+                "TopologyInfo",
+                _field("nodes", 1, _TYPE_MESSAGE, _LABEL_REPEATED, t("NUMANode")),
+            ),
+            _msg(
+                "NUMANode",
+                _field("ID", 1, _TYPE_INT64),
             ),
             _msg(
                 "PreStartContainerRequest",
@@ -167,6 +184,8 @@ RegisterRequest = _cls("RegisterRequest")
 Empty = _cls("Empty")
 ListAndWatchResponse = _cls("ListAndWatchResponse")
 Device = _cls("Device")
+TopologyInfo = _cls("TopologyInfo")
+NUMANode = _cls("NUMANode")
 PreStartContainerRequest = _cls("PreStartContainerRequest")
 PreStartContainerResponse = _cls("PreStartContainerResponse")
 AllocateRequest = _cls("AllocateRequest")

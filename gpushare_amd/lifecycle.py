@@ -55,6 +55,8 @@ class ManagerOptions:
     probe_mode: str = "subprocess"  # canary probes in a short-lived child
                                     # (daemon never maps the HIP runtime);
                                     # "inproc" keeps the r1 behavior
+    numa_topology: bool = False  # advertise Device.topology NUMA hints
+                                 # (modern kubelet TopologyManager field)
 
 
 class SharedGPUManager:
@@ -117,6 +119,7 @@ class SharedGPUManager:
             unit=self.opt.memory_unit,
             socket_dir=self.opt.socket_dir,
             allow_oversize_inventory=self.opt.allow_oversize_inventory,
+            numa_topology=self.opt.numa_topology,
         )
 
     def _start_plugin(self) -> None:

@@ -10,7 +10,11 @@
 //
 // Wire format encoded (kubelet device-plugin v1beta1, api.proto:72-88):
 //   ListAndWatchResponse{ repeated Device devices = 1 }
-//   Device{ string ID = 1; string health = 2 }
+//   Device{ string ID = 1; string health = 2; TopologyInfo topology = 3 }
+//   TopologyInfo{ repeated NUMANode nodes = 1 };  NUMANode{ int64 ID = 1 }
+// (topology is the modern upstream field — NUMA hints for the kubelet
+// TopologyManager; emitted only when a per-device numa list is given,
+// and skipped as an unknown field by pre-1.17 kubelets)
 
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
@@ -34,7 +38,9 @@ void put_varint(std::string &out, uint64_t v) {
 }
 
 // Encode one Device submessage wrapped as field 1 of ListAndWatchResponse.
-std::string encode_device(const std::string &id, const std::string &health) {
+// numa < 0 omits the topology field entirely.
+std::string encode_device(const std::string &id, const std::string &health,
+                          int64_t numa) {
   std::string dev;
   dev.push_back('\x0a');  // Device.ID, wire type 2
   put_varint(dev, id.size());
@@ -42,6 +48,18 @@ std::string encode_device(const std::string &id, const std::string &health) {
   dev.push_back('\x12');  // Device.health, wire type 2
   put_varint(dev, health.size());
   dev += health;
+  if (numa >= 0) {
+    std::string numanode;
+    numanode.push_back('\x08');  // NUMANode.ID, varint
+    put_varint(numanode, static_cast<uint64_t>(numa));
+    std::string topo;
+    topo.push_back('\x0a');  // TopologyInfo.nodes, wire type 2
+    put_varint(topo, numanode.size());
+    topo += numanode;
+    dev.push_back('\x1a');  // Device.topology, wire type 2
+    put_varint(dev, topo.size());
+    dev += topo;
+  }
 
   std::string out;
   out.push_back('\x0a');  // ListAndWatchResponse.devices, wire type 2
@@ -52,15 +70,20 @@ std::string encode_device(const std::string &id, const std::string &health) {
 
 class DeviceListCodec {
  public:
-  explicit DeviceListCodec(const std::vector<std::string> &ids) {
+  explicit DeviceListCodec(const std::vector<std::string> &ids,
+                           const std::vector<int64_t> &numa = {}) {
+    if (!numa.empty() && numa.size() != ids.size())
+      throw std::invalid_argument("numa list length != ids length");
     healthy_.reserve(ids.size());
     unhealthy_.reserve(ids.size());
     size_t total = 0;
-    for (const auto &id : ids) {
+    for (size_t i = 0; i < ids.size(); ++i) {
+      const auto &id = ids[i];
       if (id.size() > 63)
         throw std::invalid_argument("Device.ID exceeds 63 chars: " + id);
-      healthy_.push_back(encode_device(id, "Healthy"));
-      unhealthy_.push_back(encode_device(id, "Unhealthy"));
+      int64_t n = numa.empty() ? -1 : numa[i];
+      healthy_.push_back(encode_device(id, "Healthy", n));
+      unhealthy_.push_back(encode_device(id, "Unhealthy", n));
       total += healthy_.back().size();
     }
     all_healthy_.reserve(total);
@@ -96,7 +119,9 @@ class DeviceListCodec {
 PYBIND11_MODULE(_devlist, m) {
   m.doc() = "ListAndWatchResponse wire-format pre-encoder";
   py::class_<DeviceListCodec>(m, "DeviceListCodec")
-      .def(py::init<const std::vector<std::string> &>(), py::arg("ids"))
+      .def(py::init<const std::vector<std::string> &,
+                    const std::vector<int64_t> &>(),
+           py::arg("ids"), py::arg("numa") = std::vector<int64_t>{})
       .def("__len__", &DeviceListCodec::size)
       .def("encode", &DeviceListCodec::encode,
            py::arg("unhealthy") = std::vector<size_t>{});

@@ -126,6 +126,17 @@ def test_structure_matches_protoc_descriptor(golden):
     golden_shapes = _message_shapes(fdp.message_type)
     # v1beta1._build_file() is the module's own construction path
     our_shapes = _message_shapes(ours._build_file().message_type)
+
+    # documented MODERN-upstream additions (k8s >= 1.17 api.proto) absent
+    # from the vendored snapshot: Device.topology + TopologyInfo/NUMANode.
+    # Field numbers/types must match upstream exactly; everything else
+    # must match the golden descriptor byte for byte.
+    topo_field = ("topology", 3, 11, 1, "TopologyInfo")  # message, optional
+    assert topo_field in our_shapes["Device"]
+    our_shapes["Device"] = [f for f in our_shapes["Device"] if f != topo_field]
+    assert our_shapes.pop("TopologyInfo") == [("nodes", 1, 11, 3, "NUMANode")]
+    assert our_shapes.pop("NUMANode") == [("ID", 1, 3, 1, "")]  # int64
+
     assert our_shapes == golden_shapes
 
 
@@ -382,6 +393,61 @@ def test_list_and_watch_tolerates_future_topology_field():
         ("a-_-0", "Healthy"),
         ("a-_-1", "Unhealthy"),
     ]
+
+
+def test_numa_topology_encoding_first_principles():
+    """Our Device.topology emission (opt-in --numa-topology) must match
+    the modern upstream field layout exactly: TopologyInfo{nodes=1},
+    NUMANode{ID=1 varint} under Device field 3 — and decode as an
+    UNKNOWN field in the vendored-era golden classes (old-kubelet
+    safety)."""
+    from gpushare_amd.device import fakedev
+
+    ids = ["amd-x-_-0", "amd-x-_-1"]
+    payload = fakedev.encode_list_python(ids, set(), numa=[1, 1])
+    # hand-built expectation
+    numanode = _tag(1, 0) + _varint(1)
+    topo = _ld(1, numanode)
+    dev0 = _s(1, ids[0]) + _s(2, "Healthy") + _ld(3, topo)
+    dev1 = _s(1, ids[1]) + _s(2, "Healthy") + _ld(3, topo)
+    assert payload == _ld(1, dev0) + _ld(1, dev1)
+
+    # our classes decode it fully
+    resp = ours.ListAndWatchResponse.FromString(payload)
+    assert [d.topology.nodes[0].ID for d in resp.devices] == [1, 1]
+
+    # native codec agrees byte-for-byte when built
+    try:
+        from gpushare_amd import _devlist
+    except ImportError:
+        pass
+    else:
+        codec = _devlist.DeviceListCodec(ids, [1, 1])
+        assert codec.encode([]) == payload
+        # numa -1 omits topology entirely (pre-1.17 byte parity)
+        codec_off = _devlist.DeviceListCodec(ids, [-1, -1])
+        assert codec_off.encode([]) == fakedev.encode_list_python(
+            ids, set()
+        )
+
+
+def test_numa_topology_safe_for_vendored_kubelet(golden):
+    """Golden (vendored-era) classes must parse a topology-bearing list
+    losslessly with topology preserved as unknown bytes."""
+    _, classes = golden
+    from gpushare_amd.device import fakedev
+
+    payload = fakedev.encode_list_python(
+        ["a-_-0", "b-_-0"], {1}, numa=[0, 1]
+    )
+    resp = classes["ListAndWatchResponse"]()
+    resp.ParseFromString(payload)
+    assert [(d.ID, d.health) for d in resp.devices] == [
+        ("a-_-0", "Healthy"),
+        ("b-_-0", "Unhealthy"),
+    ]
+    # unknown-field preservation: re-serialization is byte-identical
+    assert resp.SerializeToString() == payload
 
 
 def test_register_request_with_unknown_field_5():

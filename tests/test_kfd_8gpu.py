@@ -263,3 +263,56 @@ class TestRealSnapshot:
         gpu = nodes[4]
         assert sorted(gpu.xgmi_peer_nodes) == [2, 3, 5, 6, 7, 8, 9]
         assert gpu.pcie_peer_nodes == [0]
+
+
+class TestNumaTopologyAdvertisement:
+    def test_plugin_advertises_numa_per_grain(self, source):
+        """--numa-topology: grains carry their GPU's NUMA domain through
+        the Device.topology field (4 GPUs per socket on the fixture)."""
+        from gpushare_amd.allocator import Allocator
+        from gpushare_amd.cluster.podmanager import PodManager
+        from gpushare_amd.deviceplugin import v1beta1 as api
+        from gpushare_amd.deviceplugin.server import GPUSharePlugin
+
+        gpus = source.devices()
+        kube = FakeKubeClient(node_name=NODE)
+        pm = PodManager(
+            kube, NODE, kubelet_client=kube.as_kubelet(), cache_ttl=0.0,
+            kubelet_retries=0, apiserver_retries=0,
+        )
+        plugin = GPUSharePlugin(
+            gpus, Allocator(gpus, pm), socket_dir="/tmp",
+            numa_topology=True,
+        )
+        resp = api.ListAndWatchResponse.FromString(
+            plugin.encoded_device_list()
+        )
+        assert len(resp.devices) == 8 * 288
+        per_gpu_numa = {}
+        for d in resp.devices:
+            gpu_idx = plugin.table.gpu_of[d.ID]
+            numa_ids = [n.ID for n in d.topology.nodes]
+            per_gpu_numa.setdefault(gpu_idx, set()).update(numa_ids)
+        # every grain of one GPU advertises exactly its GPU's NUMA node
+        for g in gpus:
+            assert per_gpu_numa[g.index] == {g.numa_node}
+
+    def test_numa_topology_off_by_default(self, source):
+        from gpushare_amd.allocator import Allocator
+        from gpushare_amd.cluster.podmanager import PodManager
+        from gpushare_amd.deviceplugin import v1beta1 as api
+        from gpushare_amd.deviceplugin.server import GPUSharePlugin
+
+        gpus = source.devices()
+        kube = FakeKubeClient(node_name=NODE)
+        pm = PodManager(
+            kube, NODE, kubelet_client=kube.as_kubelet(), cache_ttl=0.0,
+            kubelet_retries=0, apiserver_retries=0,
+        )
+        plugin = GPUSharePlugin(
+            gpus, Allocator(gpus, pm), socket_dir="/tmp"
+        )
+        resp = api.ListAndWatchResponse.FromString(
+            plugin.encoded_device_list()
+        )
+        assert all(not d.HasField("topology") for d in resp.devices)
