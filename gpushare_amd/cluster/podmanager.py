@@ -258,6 +258,7 @@ class PodManager:
             "unit": unit,
             "per_gpu_units": [g.mem_units(unit) for g in gpus],
             "xgmi": [sorted(g.xgmi_peers) for g in gpus],
+            "numa": [getattr(g, "numa_node", -1) for g in gpus],
         }
         patch = {
             "metadata": {

@@ -35,7 +35,7 @@ def discover_nodes(kube, extender) -> int:
         if not (count > 0 and total > 0):
             continue
         name = node["metadata"]["name"]
-        per_gpu, xgmi = [total // count] * count, None
+        per_gpu, xgmi, numa = [total // count] * count, None, None
         raw = (node["metadata"].get("annotations") or {}).get(
             consts.ANN_NODE_TOPOLOGY
         )
@@ -45,9 +45,10 @@ def discover_nodes(kube, extender) -> int:
                 if len(topo.get("per_gpu_units", [])) == count:
                     per_gpu = [int(u) for u in topo["per_gpu_units"]]
                     xgmi = topo.get("xgmi")
+                    numa = topo.get("numa")
             except (ValueError, TypeError, KeyError) as e:
                 log.warning("bad topology annotation on %s: %s", name, e)
-        extender.register_node(name, per_gpu, xgmi=xgmi)
+        extender.register_node(name, per_gpu, xgmi=xgmi, numa=numa)
         n += 1
     return n
 

@@ -30,6 +30,7 @@ class NodeGPUState:
     per_gpu_units: list[int]                 # capacity per GPU
     allocated: list[int] = field(default_factory=list)
     xgmi: list[tuple] = field(default_factory=list)   # adjacency per GPU
+    numa: list[int] = field(default_factory=list)     # NUMA node per GPU (-1 unknown)
 
     def __post_init__(self):
         if not self.allocated:
@@ -37,6 +38,10 @@ class NodeGPUState:
         if len(self.xgmi) != len(self.per_gpu_units):
             self.xgmi = [tuple(p) for p in self.xgmi] + [()] * (
                 len(self.per_gpu_units) - len(self.xgmi)
+            )
+        if len(self.numa) != len(self.per_gpu_units):
+            self.numa = list(self.numa) + [-1] * (
+                len(self.per_gpu_units) - len(self.numa)
             )
 
     def free(self, idx: int) -> int:
@@ -63,6 +68,15 @@ class NodeGPUState:
         s = set(idxs)
         return sum(1 for i in idxs for p in self.xgmi[i] if p in s) // 2
 
+    def _numa_domains(self, idxs: tuple) -> int:
+        """Distinct NUMA domains a GPU set spans (unknown counts as its
+        own domain, so an all-unknown node is unaffected).  On an 8-OAM
+        full-mesh MI355X every pair ties on xGMI edges — host-memory
+        locality (4 GPUs per socket) is the honest second key."""
+        return len({
+            self.numa[i] if self.numa[i] >= 0 else -(i + 2) for i in idxs
+        })
+
     def best_fit_multi(
         self, request: int, spread: bool = False
     ) -> Optional[dict[int, int]]:
@@ -82,11 +96,16 @@ class NodeGPUState:
                 total = sum(frees[i] for i in combo)
                 if total < request:
                     continue
-                key = (-self._xgmi_edges(combo), total, combo)
+                key = (
+                    -self._xgmi_edges(combo),
+                    self._numa_domains(combo),
+                    total,
+                    combo,
+                )
                 if best is None or key < best:
                     best = key
             if best is not None:
-                combo = best[2]
+                combo = best[-1]
                 # fill least-free first; the last GPU takes the remainder
                 remaining = request
                 split: dict[int, int] = {}
@@ -117,13 +136,15 @@ class BinpackState:
 
     def set_node(self, node: str, per_gpu_units: list[int],
                  allocated: Optional[list[int]] = None,
-                 xgmi: Optional[list] = None) -> None:
+                 xgmi: Optional[list] = None,
+                 numa: Optional[list] = None) -> None:
         with self._lock:
             self.nodes[node] = NodeGPUState(
                 node,
                 list(per_gpu_units),
                 list(allocated or []),
                 [tuple(p) for p in (xgmi or [])],
+                [int(n) for n in (numa or [])],
             )
 
     def filter_nodes(self, request: int, candidates: list[str]) -> list[str]:

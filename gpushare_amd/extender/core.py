@@ -54,8 +54,9 @@ class GPUShareExtender:
     # state sync
     # ------------------------------------------------------------------ #
     def register_node(self, node: str, per_gpu_units: list[int],
-                      xgmi: Optional[list] = None) -> None:
-        self.state.set_node(node, per_gpu_units, xgmi=xgmi)
+                      xgmi: Optional[list] = None,
+                      numa: Optional[list] = None) -> None:
+        self.state.set_node(node, per_gpu_units, xgmi=xgmi, numa=numa)
 
     def resync(self, nodes: Optional[list[str]] = None) -> None:
         """Rebuild the ledger from pod annotations (source of truth)."""
@@ -86,7 +87,9 @@ class GPUShareExtender:
                 idx = podutils.gpu_id_from_annotation(pod)
                 if 0 <= idx < len(allocated):
                     allocated[idx] += mem
-            self.state.set_node(node, st.per_gpu_units, allocated, st.xgmi)
+            self.state.set_node(
+                node, st.per_gpu_units, allocated, st.xgmi, numa=st.numa
+            )
         live = {
             (podutils.pod_namespace(p), podutils.pod_name(p)) for p in pods
         }

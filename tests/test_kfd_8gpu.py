@@ -316,3 +316,61 @@ class TestNumaTopologyAdvertisement:
             plugin.encoded_device_list()
         )
         assert all(not d.HasField("topology") for d in resp.devices)
+
+
+class TestNumaTiebreak:
+    """On an 8-OAM full xGMI mesh every GPU pair ties on link count; the
+    second placement key is host-memory locality (4 GPUs per socket)."""
+
+    def test_split_prefers_same_socket(self, source):
+        gpus = source.devices()
+        state = NodeGPUState(
+            NODE,
+            [g.mem_units("GiB") for g in gpus],
+            xgmi=[g.xgmi_peers for g in gpus],
+            numa=[g.numa_node for g in gpus],
+        )
+        # equal free everywhere -> edges and free tie across all pairs;
+        # numa must break the tie: the chosen pair shares a socket
+        for i in range(8):
+            state.allocated[i] = 200
+        split = state.best_fit_multi(170)
+        assert len(split) == 2
+        domains = {state.numa[i] for i in split}
+        assert len(domains) == 1, (
+            f"split {sorted(split)} spans sockets {domains}"
+        )
+
+    def test_cross_socket_still_allowed_when_forced(self, source):
+        gpus = source.devices()
+        numa = [g.numa_node for g in gpus]
+        state = NodeGPUState(
+            NODE,
+            [g.mem_units("GiB") for g in gpus],
+            xgmi=[g.xgmi_peers for g in gpus],
+            numa=numa,
+        )
+        # leave free memory on exactly one GPU per socket: a 2-GPU split
+        # must go cross-socket rather than fail
+        socket0 = next(i for i in range(8) if numa[i] == 0)
+        socket1 = next(i for i in range(8) if numa[i] == 1)
+        for i in range(8):
+            state.allocated[i] = 288 if i not in (socket0, socket1) else 188
+        split = state.best_fit_multi(180)
+        assert split is not None
+        assert sorted(split) == sorted([socket0, socket1])
+
+    def test_unknown_numa_neutral(self):
+        """All-unknown NUMA (old plugins, missing annotation) must not
+        change pre-r2 behavior: tightest pack still wins."""
+        state = NodeGPUState(
+            "n", [16, 16, 16], allocated=[10, 8, 8],
+            xgmi=[(1, 2), (0, 2), (0, 1)],
+        )
+        split = state.best_fit_multi(10)
+        # no single fits? 16-10=6,8,8 free; 10 needs 2 GPUs; tightest
+        # total picks the two 8-free GPUs (16 total) over 6+8=14? No:
+        # smaller total wins -> {0,1} or {0,2} with 14. Edges all equal.
+        assert len(split) == 2
+        assert sum(split.values()) == 10
+        assert 0 in split  # the 6-free GPU is in the tightest pair
