@@ -681,3 +681,36 @@ def test_gpushare_top_on_real_gpu(source):
     assert top.main([], source=source, out=out) == 0
     text = out.getvalue()
     assert "GiB" in text and "renderD" in text
+
+
+@pytest.mark.gpu
+def test_numa_topology_advertised_from_real_kfd(source, tmp_path):
+    """--numa-topology on the real box: every grain must carry its GPU's
+    actual NUMA domain (from the DRM device's numa_node) through the
+    modern Device.topology field."""
+    from gpushare_amd import consts
+    from gpushare_amd.allocator import Allocator
+    from gpushare_amd.cluster.kubeclient import FakeKubeClient
+    from gpushare_amd.cluster.podmanager import PodManager
+    from gpushare_amd.deviceplugin import v1beta1 as api
+    from gpushare_amd.deviceplugin.server import GPUSharePlugin
+
+    gpus = source.devices()
+    kube = FakeKubeClient(node_name="gpu-node")
+    pm = PodManager(
+        kube, "gpu-node", kubelet_client=kube.as_kubelet(), cache_ttl=0.0
+    )
+    plugin = GPUSharePlugin(
+        gpus,
+        Allocator(gpus, pm),
+        socket_dir=str(tmp_path),
+        numa_topology=True,
+    )
+    resp = api.ListAndWatchResponse.FromString(plugin.encoded_device_list())
+    assert len(resp.devices) == sum(g.mem_units(consts.GIB) for g in gpus)
+    for d in resp.devices:
+        gpu = gpus[plugin.table.gpu_of[d.ID]]
+        if gpu.numa_node >= 0:
+            assert [n.ID for n in d.topology.nodes] == [gpu.numa_node]
+        else:
+            assert not d.HasField("topology")

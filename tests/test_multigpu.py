@@ -331,6 +331,12 @@ class TestTopologyAnnotation:
         st = ext.state.nodes[NODE]
         assert st.per_gpu_units == [16, 16, 16, 16]
         assert st.xgmi[0] == (1, 2, 3)
+        # r2: per-GPU NUMA nodes ride the same annotation end to end
+        assert "numa" in topo
+        assert st.numa == [g.numa_node for g in gpus]
+        # resync must not drop the numa list (r2 regression guard)
+        ext.resync()
+        assert ext.state.nodes[NODE].numa == st.numa
 
 
 class TestAutoRelease:
