@@ -322,6 +322,21 @@ def main():
         allocated += a
         failed += f
         packing_samples.append(peak)
+        # stderr heartbeat so a soak killed by an outer timeout still
+        # leaves rate/latency evidence in its log (never on stdout — the
+        # one-JSON-line contract stays intact)
+        if rank == 0 and step and step % 2000 == 0:
+            dt = time.perf_counter() - t0
+            lat = sorted(lat_allocate)
+            p50 = lat[len(lat) // 2] * 1e3 if lat else 0.0
+            print(
+                f"# progress step={step}/{args.steps} "
+                f"pods={allocated} failed={failed} "
+                f"rate={allocated / dt:.0f}/s/rank0 "
+                f"alloc_p50={p50:.2f}ms",
+                file=sys.stderr,
+                flush=True,
+            )
     if distributed:
         dist.barrier()
     if torch.cuda.is_available():
