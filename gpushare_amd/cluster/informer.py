@@ -68,6 +68,7 @@ class PodInformer:
         # only bounds memory, not correctness).
         self._tombstones: dict[str, tuple[int, float]] = {}
         self._tombstone_ttl = 600.0
+        self._tombstone_cap = 10_000
         self._lock = threading.Lock()
         self._cv = threading.Condition(self._lock)
         self._version = 0            # bumped on every applied event / relist
@@ -228,13 +229,23 @@ class PodInformer:
                         max(_rv(pod), _rv(cur) if cur else 0),
                         now,
                     )
-                    if len(self._tombstones) > 10_000:
-                        cutoff = now - self._tombstone_ttl
-                        self._tombstones = {
-                            u: (rv, t)
-                            for u, (rv, t) in self._tombstones.items()
-                            if t > cutoff
-                        }
+                    # evict from the FRONT (dict insertion order == time
+                    # order): age-expired entries, plus a hard size cap —
+                    # O(evicted) per event, NEVER a full-dict rebuild (a
+                    # rebuild per DELETED melted the informer thread in
+                    # the 256k-pod soak: >10k young tombstones meant every
+                    # delete paid O(n) for zero evictions)
+                    cutoff = now - self._tombstone_ttl
+                    while self._tombstones:
+                        first = next(iter(self._tombstones))
+                        _rv0, t0 = self._tombstones[first]
+                        if (
+                            t0 <= cutoff
+                            or len(self._tombstones) > self._tombstone_cap
+                        ):
+                            del self._tombstones[first]
+                        else:
+                            break
             elif etype in ("ADDED", "MODIFIED"):
                 tomb = self._tombstones.get(uid)
                 if tomb is not None and _rv(pod) <= tomb[0]:

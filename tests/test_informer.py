@@ -499,3 +499,46 @@ class TestAdversarialInterleavings:
         finally:
             inf.stop()
             kube.stream.put(None)
+
+
+class TestTombstoneScaling:
+    def test_100k_deletes_stay_amortized_and_capped(self):
+        """Soak regression (256k-pod run): tombstone eviction must be
+        O(evicted) per DELETED — a full-dict rebuild per event melted the
+        informer thread once >10k young tombstones accumulated.  100k
+        unique deletes must stay fast and the table capped."""
+        inf = PodInformer(kube=None, node_name=NODE)
+        t0 = time.perf_counter()
+        for i in range(100_000):
+            pod = make_pod(f"d{i}", 4, node=NODE)
+            pod["metadata"]["uid"] = f"uid-soak-{i}"
+            pod["metadata"]["resourceVersion"] = str(i + 1)
+            inf._apply("ADDED", pod)
+            inf._apply("DELETED", pod)
+        elapsed = time.perf_counter() - t0
+        assert len(inf._tombstones) <= inf._tombstone_cap + 1
+        assert inf.pods() == []
+        # generous bound: the O(n)-rebuild version takes minutes here
+        assert elapsed < 10.0, f"tombstone path too slow: {elapsed:.1f}s"
+
+    def test_eviction_never_breaks_recent_protection(self):
+        """The most RECENT deletions (the ones a stale re-list could
+        resurrect) must survive cap eviction."""
+        inf = PodInformer(kube=None, node_name=NODE)
+        inf._tombstone_cap = 100
+        for i in range(500):
+            pod = make_pod(f"e{i}", 4, node=NODE)
+            pod["metadata"]["uid"] = f"uid-cap-{i}"
+            pod["metadata"]["resourceVersion"] = str(i + 1)
+            inf._apply("ADDED", pod)
+            inf._apply("DELETED", pod)
+        assert len(inf._tombstones) <= 101
+        # the newest tombstones are retained
+        assert "uid-cap-499" in inf._tombstones
+        assert "uid-cap-0" not in inf._tombstones
+        # and still block a stale re-list of a recent delete
+        stale = make_pod("e499", 4, node=NODE)
+        stale["metadata"]["uid"] = "uid-cap-499"
+        stale["metadata"]["resourceVersion"] = "450"
+        inf._merge_snapshot([stale])
+        assert inf.pods() == []
