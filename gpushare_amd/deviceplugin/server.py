@@ -129,7 +129,10 @@ class GPUSharePlugin:
     # RPC behaviors
     # ------------------------------------------------------------------ #
     def _get_options(self, request, context) -> "api.DevicePluginOptions":
-        return api.DevicePluginOptions(pre_start_required=False)
+        return api.DevicePluginOptions(
+            pre_start_required=False,
+            get_preferred_allocation_available=True,
+        )
 
     def _list_and_watch(self, request, context):
         """Stream: full list once, then re-send on every health change.
@@ -163,6 +166,51 @@ class GPUSharePlugin:
     def _allocate(self, request, context) -> "api.AllocateResponse":
         return self.allocator.allocate(request)
 
+    def _get_preferred_allocation(
+        self, request, context
+    ) -> "api.PreferredAllocationResponse":
+        """Modern kubelet (>= 1.19) asks which grains to hand Allocate.
+
+        Grains are equivalent WITHIN a GPU but not across GPUs: an
+        arbitrary kubelet pick can strand availability across several
+        GPUs while the extender packs by whole-GPU memory.  Answer
+        binpack-consistently: after honoring must_include, fill from the
+        single GPU that can satisfy the remainder with the FEWEST
+        available grains (tightest fit), falling back to most-available
+        order across GPUs for oversize requests."""
+        resp = api.PreferredAllocationResponse()
+        for cr in request.container_requests:
+            out = resp.container_responses.add()
+            chosen = list(cr.must_include_deviceIDs)
+            need = cr.allocation_size - len(chosen)
+            if need < 0:
+                out.deviceIDs.extend(chosen[: cr.allocation_size])
+                continue
+            chosen_set = set(chosen)
+            by_gpu: dict[int, list[str]] = {}
+            for dev_id in cr.available_deviceIDs:
+                if dev_id in chosen_set:
+                    continue
+                gpu_idx = self.table.gpu_of.get(dev_id)
+                if gpu_idx is not None:
+                    by_gpu.setdefault(gpu_idx, []).append(dev_id)
+            # tightest single GPU first (binpack), most-available fallback
+            fitting = sorted(
+                (ids for ids in by_gpu.values() if len(ids) >= need),
+                key=len,
+            )
+            if need and fitting:
+                chosen.extend(fitting[0][:need])
+            else:
+                for ids in sorted(by_gpu.values(), key=len, reverse=True):
+                    if need <= 0:
+                        break
+                    take = ids[:need]
+                    chosen.extend(take)
+                    need -= len(take)
+            out.deviceIDs.extend(chosen)
+        return resp
+
     def _pre_start(self, request, context) -> "api.PreStartContainerResponse":
         return api.PreStartContainerResponse()
 
@@ -195,6 +243,13 @@ class GPUSharePlugin:
             "PreStartContainer": grpc.unary_unary_rpc_method_handler(
                 self._pre_start,
                 request_deserializer=api.PreStartContainerRequest.FromString,
+                response_serializer=lambda m: m.SerializeToString(),
+            ),
+            "GetPreferredAllocation": grpc.unary_unary_rpc_method_handler(
+                self._get_preferred_allocation,
+                request_deserializer=(
+                    api.PreferredAllocationRequest.FromString
+                ),
                 response_serializer=lambda m: m.SerializeToString(),
             ),
         }

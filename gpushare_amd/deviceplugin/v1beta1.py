@@ -60,9 +60,14 @@ def _build_file() -> descriptor_pb2.FileDescriptorProto:
 
     f.message_type.extend(
         [
+            # pre_start_required is the vendored-era field; field 2 and the
+            # GetPreferredAllocation messages below are the MODERN upstream
+            # additions (k8s >= 1.19 api.proto) — a pre-1.19 kubelet skips
+            # the unknown bool and never calls the RPC.
             _msg(
                 "DevicePluginOptions",
                 _field("pre_start_required", 1, _TYPE_BOOL),
+                _field("get_preferred_allocation_available", 2, _TYPE_BOOL),
             ),
             _msg(
                 "RegisterRequest",
@@ -153,6 +158,42 @@ def _build_file() -> descriptor_pb2.FileDescriptorProto:
                 nested=(_map_entry("EnvsEntry"), _map_entry("AnnotationsEntry")),
             ),
             _msg(
+                "PreferredAllocationRequest",
+                _field(
+                    "container_requests",
+                    1,
+                    _TYPE_MESSAGE,
+                    _LABEL_REPEATED,
+                    t("ContainerPreferredAllocationRequest"),
+                ),
+            ),
+            _msg(
+                "ContainerPreferredAllocationRequest",
+                _field("available_deviceIDs", 1, _TYPE_STRING, _LABEL_REPEATED),
+                _field(
+                    "must_include_deviceIDs", 2, _TYPE_STRING, _LABEL_REPEATED
+                ),
+                _field(
+                    "allocation_size",
+                    3,
+                    descriptor_pb2.FieldDescriptorProto.TYPE_INT32,
+                ),
+            ),
+            _msg(
+                "PreferredAllocationResponse",
+                _field(
+                    "container_responses",
+                    1,
+                    _TYPE_MESSAGE,
+                    _LABEL_REPEATED,
+                    t("ContainerPreferredAllocationResponse"),
+                ),
+            ),
+            _msg(
+                "ContainerPreferredAllocationResponse",
+                _field("deviceIDs", 1, _TYPE_STRING, _LABEL_REPEATED),
+            ),
+            _msg(
                 "Mount",
                 _field("container_path", 1, _TYPE_STRING),
                 _field("host_path", 2, _TYPE_STRING),
@@ -190,6 +231,14 @@ PreStartContainerRequest = _cls("PreStartContainerRequest")
 PreStartContainerResponse = _cls("PreStartContainerResponse")
 AllocateRequest = _cls("AllocateRequest")
 ContainerAllocateRequest = _cls("ContainerAllocateRequest")
+PreferredAllocationRequest = _cls("PreferredAllocationRequest")
+ContainerPreferredAllocationRequest = _cls(
+    "ContainerPreferredAllocationRequest"
+)
+PreferredAllocationResponse = _cls("PreferredAllocationResponse")
+ContainerPreferredAllocationResponse = _cls(
+    "ContainerPreferredAllocationResponse"
+)
 AllocateResponse = _cls("AllocateResponse")
 ContainerAllocateResponse = _cls("ContainerAllocateResponse")
 Mount = _cls("Mount")
@@ -204,3 +253,4 @@ METHOD_GET_OPTIONS = f"/{DEVICEPLUGIN_SERVICE}/GetDevicePluginOptions"
 METHOD_LIST_AND_WATCH = f"/{DEVICEPLUGIN_SERVICE}/ListAndWatch"
 METHOD_ALLOCATE = f"/{DEVICEPLUGIN_SERVICE}/Allocate"
 METHOD_PRE_START = f"/{DEVICEPLUGIN_SERVICE}/PreStartContainer"
+METHOD_GET_PREFERRED = f"/{DEVICEPLUGIN_SERVICE}/GetPreferredAllocation"

@@ -183,3 +183,61 @@ def test_pre_start_container_rpc(tmp_socket_dir):
     finally:
         plugin.stop()
         kubelet.stop()
+
+
+def test_get_preferred_allocation_binpacks_one_gpu(harness):
+    """Modern kubelet RPC (k8s >= 1.19): preferred grains must come from
+    ONE GPU — the tightest-fitting one — so kubelet grain bookkeeping
+    matches the extender's whole-GPU packing."""
+    import grpc
+
+    from gpushare_amd.deviceplugin import v1beta1 as api
+
+    kube, plugin, kubelet = harness("2x8GiB")
+    client = kubelet.wait_for_plugin(consts.RESOURCE_NAME)
+    devices = client.wait_for_devices(min_count=16)
+
+    opts = client.options
+    assert opts.get_preferred_allocation_available is True
+
+    channel = grpc.insecure_channel(f"unix://{plugin.socket_path}")
+    call = channel.unary_unary(
+        api.METHOD_GET_PREFERRED,
+        request_serializer=lambda m: m.SerializeToString(),
+        response_deserializer=api.PreferredAllocationResponse.FromString,
+    )
+    try:
+        # GPU0 has 3 grains available, GPU1 has 8: a request for 3 must
+        # take the tightest fit (all of GPU0's remainder)
+        gpu0 = sorted(i for i in devices if plugin.table.gpu_of[i] == 0)
+        gpu1 = sorted(i for i in devices if plugin.table.gpu_of[i] == 1)
+        req = api.PreferredAllocationRequest()
+        cr = req.container_requests.add()
+        cr.available_deviceIDs.extend(gpu0[:3] + gpu1)
+        cr.allocation_size = 3
+        resp = call(req, timeout=5)
+        chosen = list(resp.container_responses[0].deviceIDs)
+        assert len(chosen) == 3
+        assert {plugin.table.gpu_of[i] for i in chosen} == {0}
+
+        # must_include is honored and the remainder stays on one GPU
+        req2 = api.PreferredAllocationRequest()
+        cr2 = req2.container_requests.add()
+        cr2.available_deviceIDs.extend(gpu0 + gpu1)
+        cr2.must_include_deviceIDs.append(gpu1[0])
+        cr2.allocation_size = 4
+        resp2 = call(req2, timeout=5)
+        chosen2 = list(resp2.container_responses[0].deviceIDs)
+        assert len(chosen2) == 4
+        assert gpu1[0] in chosen2
+
+        # oversize (no single GPU fits): spread, most-available first,
+        # full count still returned
+        req3 = api.PreferredAllocationRequest()
+        cr3 = req3.container_requests.add()
+        cr3.available_deviceIDs.extend(gpu0[:2] + gpu1[:3])
+        cr3.allocation_size = 5
+        resp3 = call(req3, timeout=5)
+        assert len(resp3.container_responses[0].deviceIDs) == 5
+    finally:
+        channel.close()

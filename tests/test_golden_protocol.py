@@ -127,8 +127,9 @@ def test_structure_matches_protoc_descriptor(golden):
     # v1beta1._build_file() is the module's own construction path
     our_shapes = _message_shapes(ours._build_file().message_type)
 
-    # documented MODERN-upstream additions (k8s >= 1.17 api.proto) absent
-    # from the vendored snapshot: Device.topology + TopologyInfo/NUMANode.
+    # documented MODERN-upstream additions absent from the vendored
+    # snapshot: Device.topology + TopologyInfo/NUMANode (k8s >= 1.17) and
+    # GetPreferredAllocation + DevicePluginOptions field 2 (k8s >= 1.19).
     # Field numbers/types must match upstream exactly; everything else
     # must match the golden descriptor byte for byte.
     topo_field = ("topology", 3, 11, 1, "TopologyInfo")  # message, optional
@@ -136,6 +137,32 @@ def test_structure_matches_protoc_descriptor(golden):
     our_shapes["Device"] = [f for f in our_shapes["Device"] if f != topo_field]
     assert our_shapes.pop("TopologyInfo") == [("nodes", 1, 11, 3, "NUMANode")]
     assert our_shapes.pop("NUMANode") == [("ID", 1, 3, 1, "")]  # int64
+
+    gpa_field = ("get_preferred_allocation_available", 2, 8, 1, "")  # bool
+    assert gpa_field in our_shapes["DevicePluginOptions"]
+    our_shapes["DevicePluginOptions"] = [
+        f for f in our_shapes["DevicePluginOptions"] if f != gpa_field
+    ]
+    assert our_shapes.pop("PreferredAllocationRequest") == [
+        ("container_requests", 1, 11, 3, "ContainerPreferredAllocationRequest")
+    ]
+    assert our_shapes.pop("ContainerPreferredAllocationRequest") == sorted([
+        ("available_deviceIDs", 1, 9, 3, ""),
+        ("must_include_deviceIDs", 2, 9, 3, ""),
+        ("allocation_size", 3, 5, 1, ""),  # int32
+    ])
+    assert our_shapes.pop("PreferredAllocationResponse") == [
+        (
+            "container_responses",
+            1,
+            11,
+            3,
+            "ContainerPreferredAllocationResponse",
+        )
+    ]
+    assert our_shapes.pop("ContainerPreferredAllocationResponse") == [
+        ("deviceIDs", 1, 9, 3, "")
+    ]
 
     assert our_shapes == golden_shapes
 
