@@ -38,13 +38,13 @@ from gpushare_amd.deviceplugin.stubkubelet import StubKubelet
 
 
 def measure(spec: str, unit: str, stream: bool = True,
-            client_max_mb: int = 4) -> dict:
+            client_max_mb: int = 4, numa: bool = False) -> dict:
     gpus = MockSource.from_spec(spec).devices()
     t0 = time.perf_counter()
     table = FakeDeviceTable.build(gpus, unit)
     t_table = time.perf_counter() - t0
     t0 = time.perf_counter()
-    codec = make_codec(table.ids)
+    codec = make_codec(table.ids, numa=table.numa_of if numa else None)
     t_codec = time.perf_counter() - t0
     t0 = time.perf_counter()
     payload = codec.encode([])
@@ -57,6 +57,7 @@ def measure(spec: str, unit: str, stream: bool = True,
     out = {
         "spec": spec,
         "unit": unit,
+        "numa_topology": numa,
         "devices": len(table),
         "payload_mb": round(len(payload) / 1e6, 2),
         "table_build_ms": round(t_table * 1e3, 1),
@@ -107,6 +108,8 @@ def main() -> int:
     args = p.parse_args()
     results = [
         measure("8x288GiB", consts.GIB),
+        # modern Device.topology NUMA hints: quantify the payload cost
+        measure("8x288GiB", consts.GIB, numa=True),
         # MiB grain exceeds the stock kubelet 4 MiB gRPC limit (the plugin
         # refuses it by default); measured here with a raised client limit
         measure("1x288GiB", consts.MIB, client_max_mb=64),
