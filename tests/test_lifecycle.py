@@ -293,3 +293,42 @@ def test_probe_in_subprocess_fails_loudly_without_gpu():
         _pytest.skip("GPU present; covered by the gpu-marked test")
     with _pytest.raises(RuntimeError, match="canary subprocess"):
         health_mod.probe_in_subprocess(0, vram_probe_mb=1)
+
+
+def test_health_monitor_probe_timeout_flips_unhealthy(
+    tmp_socket_dir, monkeypatch
+):
+    """A probe child that WEDGES (TimeoutExpired) must flip the GPU
+    Unhealthy — a hung GPU is exactly when the canary must speak."""
+    import subprocess as sp
+
+    from gpushare_amd import health as health_mod
+    from gpushare_amd.allocator import Allocator
+    from gpushare_amd.cluster.podmanager import PodManager
+    from gpushare_amd.deviceplugin.server import GPUSharePlugin
+
+    kube = FakeKubeClient("node-a")
+    source = MockSource.from_spec("1x8GiB")
+    gpus = source.devices()
+    pm = PodManager(
+        kube, "node-a", kubelet_client=kube.as_kubelet(), cache_ttl=0
+    )
+    plugin = GPUSharePlugin(
+        gpus, Allocator(gpus, pm), socket_dir=tmp_socket_dir
+    )
+
+    def wedged_probe(gpu_index, vram_probe_mb=32, timeout=60.0):
+        raise sp.TimeoutExpired(cmd="canary", timeout=timeout)
+
+    monkeypatch.setattr(health_mod, "probe_in_subprocess", wedged_probe)
+    mon = HealthMonitor(
+        source, plugin, deep_probe_interval=0.02, probe_mode="subprocess"
+    )
+    mon.start()
+    try:
+        deadline = time.monotonic() + 5
+        while plugin._unhealthy_gpus != {0} and time.monotonic() < deadline:
+            time.sleep(0.01)
+        assert plugin._unhealthy_gpus == {0}
+    finally:
+        mon.stop()
