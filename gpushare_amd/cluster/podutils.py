@@ -107,7 +107,14 @@ def gpu_split_from_pod(pod: dict) -> Optional[dict[int, int]]:
     try:
         for per_gpu in alloc.values():
             for idx_s, units in per_gpu.items():
-                merged[int(idx_s)] = merged.get(int(idx_s), 0) + int(units)
+                idx, u = int(idx_s), int(units)
+                # annotations are cluster data any pod-patcher can write:
+                # reject absurd indices/units instead of letting them
+                # poison the extender's resync ledger (1 Mi units = 1 PiB
+                # at GiB grain — far beyond any node)
+                if not (0 <= idx < 1024 and 0 < u <= 1 << 20):
+                    return None
+                merged[idx] = merged.get(idx, 0) + u
     except (TypeError, ValueError, AttributeError):
         return None
     return merged or None

@@ -81,3 +81,120 @@ def test_allocation_map_annotation():
     assert podutils.allocation_map_from_annotation(pod) == {"c0": {"2": 4}}
     pod["metadata"]["annotations"][consts.ANN_GPUSHARE_ALLOCATION] = "not-json"
     assert podutils.allocation_map_from_annotation(pod) is None
+
+
+class TestAdversarialAnnotations:
+    """Annotations are tenant-influenced cluster data: every parser must
+    degrade safely on garbage (hypothesis fuzz + known nasty cases)."""
+
+    NASTY = [
+        None, "", " ", "NaN", "1e309", "-0", "0x10", "۱۲۳",  # non-ASCII digits
+        "9" * 400, "[]", "{}", '{"a":', "\x00", "true", "null", "-1",
+    ]
+
+    def test_gpu_id_never_raises(self):
+        for val in self.NASTY:
+            pod = {"metadata": {"annotations": {}}}
+            if val is not None:
+                pod["metadata"]["annotations"][
+                    consts.ENV_RESOURCE_INDEX
+                ] = val
+            idx = podutils.gpu_id_from_annotation(pod)
+            assert isinstance(idx, int)
+
+    def test_assume_time_never_raises(self):
+        for val in self.NASTY:
+            pod = {"metadata": {"annotations": {}}}
+            if val is not None:
+                pod["metadata"]["annotations"][
+                    consts.ENV_RESOURCE_ASSUME_TIME
+                ] = val
+            t = podutils.assume_time_from_annotation(pod)
+            assert isinstance(t, int)
+
+    def test_allocation_map_never_raises(self):
+        for val in self.NASTY + [
+            '{"c": "notdict"}',
+            '{"c": {"x": 1}}',           # non-int GPU key
+            '{"c": {"0": "NaN"}}',       # non-int units
+            '{"c": {"0": [1]}}',
+            '[1,2,3]',
+            '{"c": {"0": 1e99}}',
+        ]:
+            pod = {"metadata": {"annotations": {}}}
+            if val is not None:
+                pod["metadata"]["annotations"][
+                    consts.ANN_GPUSHARE_ALLOCATION
+                ] = val
+            split = podutils.gpu_split_from_pod(pod)
+            assert split is None or (
+                isinstance(split, dict)
+                and all(
+                    isinstance(k, int) and isinstance(v, int)
+                    for k, v in split.items()
+                )
+            )
+
+    def test_is_assumed_never_raises_on_garbage(self):
+        for val in self.NASTY:
+            pod = {
+                "metadata": {
+                    "annotations": {
+                        consts.ENV_RESOURCE_ASSUME_TIME: val or "",
+                        consts.ENV_ASSIGNED_FLAG: val or "",
+                    }
+                },
+                "spec": {
+                    "containers": [
+                        {"resources": {"limits": {consts.RESOURCE_NAME: "4"}}}
+                    ]
+                },
+                "status": {"phase": "Pending"},
+            }
+            assert podutils.is_assumed_pod(pod) in (True, False)
+
+    def test_hypothesis_fuzz_annotations(self):
+        try:
+            from hypothesis import given, settings, strategies as st
+        except ImportError:
+            import pytest
+
+            pytest.skip("hypothesis unavailable")
+
+        @settings(max_examples=300, deadline=None)
+        @given(st.text(max_size=80))
+        def run(val):
+            pod = {
+                "metadata": {
+                    "annotations": {
+                        consts.ENV_RESOURCE_INDEX: val,
+                        consts.ENV_RESOURCE_ASSUME_TIME: val,
+                        consts.ANN_GPUSHARE_ALLOCATION: val,
+                    }
+                }
+            }
+            podutils.gpu_id_from_annotation(pod)
+            podutils.assume_time_from_annotation(pod)
+            podutils.gpu_split_from_pod(pod)
+
+        run()
+
+
+def test_split_rejects_absurd_indices_and_units():
+    from helpers import make_pod
+    import json as _json
+
+    for bad in (
+        {"c": {"0": 10**19}},          # poisoned units
+        {"c": {"0": 0}},               # zero units
+        {"c": {"0": -4}},              # negative
+        {"c": {"99999": 4}},           # absurd GPU index
+        {"c": {"-1": 4}},
+    ):
+        pod = make_pod(
+            "p", 4,
+            extra_annotations={
+                consts.ANN_GPUSHARE_ALLOCATION: _json.dumps(bad)
+            },
+        )
+        assert podutils.gpu_split_from_pod(pod) is None, bad
