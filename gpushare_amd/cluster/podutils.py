@@ -24,7 +24,12 @@ from .. import consts
 # --------------------------------------------------------------------------- #
 
 def annotations(pod: dict) -> dict:
-    return pod.get("metadata", {}).get("annotations") or {}
+    # pods are untyped JSON from /pods or the apiserver — a malformed
+    # payload must degrade to "no annotations", not an AttributeError in
+    # every caller
+    md = pod.get("metadata", {})
+    anns = md.get("annotations") if isinstance(md, dict) else None
+    return anns if isinstance(anns, dict) else {}
 
 
 def pod_name(pod: dict) -> str:

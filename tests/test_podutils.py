@@ -198,3 +198,22 @@ def test_split_rejects_absurd_indices_and_units():
             },
         )
         assert podutils.gpu_split_from_pod(pod) is None, bad
+
+
+def test_malformed_pod_shapes_degrade_cleanly():
+    """/pods is untyped JSON; structurally-broken pods must parse as
+    'not a gpushare pod', never crash the annotation accessors."""
+    shapes = [
+        {},
+        {"metadata": None},
+        {"metadata": []},
+        {"metadata": "x"},
+        {"metadata": {"annotations": []}},
+        {"metadata": {"annotations": "notdict"}},
+        {"metadata": {"annotations": None}},
+    ]
+    for pod in shapes:
+        assert podutils.annotations(pod) == {}
+        assert podutils.gpu_id_from_annotation(pod) == -1
+        assert podutils.assume_time_from_annotation(pod) == 0
+        assert podutils.gpu_split_from_pod(pod) is None
