@@ -136,7 +136,7 @@ def test_mock_heterogeneous_spec():
 def test_memguard_lib_loads_and_reads_env(tmp_path):
     """CPU-side smoke of libgpushare_memguard.so: the lib must dlopen
     anywhere (links only libdl/libc), parse the env budget, and map its
-    pod-scoped slot table (no HIP runtime involved)."""
+    container-scoped slot table (no HIP runtime involved)."""
     import os
     import subprocess
     import sys
