@@ -249,10 +249,60 @@ def build_node_infos(kube, node_name: str = "") -> list[NodeInfo]:
     return infos
 
 
+def display_json(infos: list[NodeInfo], out=sys.stdout) -> None:
+    """Machine-readable dump (no reference counterpart — the reference's
+    tabwriter output is awkward to script against)."""
+    import json
+
+    nodes = []
+    for n in infos:
+        devs = {}
+        for idx, dev in sorted(n.devs.items()):
+            devs[str(idx)] = {
+                "total": n.gpu_total(idx) if idx >= 0 else None,
+                "used": dev["used"],
+                "pods": [
+                    {
+                        "namespace": podutils.pod_namespace(p),
+                        "name": podutils.pod_name(p),
+                        "gpu_mem": get_allocation(p).get(idx, 0),
+                    }
+                    for p in dev["pods"]
+                ],
+            }
+        nodes.append(
+            {
+                "name": n.name,
+                "address": n.address,
+                "gpu_count": n.gpu_count,
+                "gpu_mem_total": n.total_mem,
+                "gpu_mem_used": n.used_mem,
+                "per_gpu_units": n.per_gpu_units or None,
+                "devices": devs,
+            }
+        )
+    json.dump(
+        {
+            "unit": infer_memory_unit(infos),
+            "nodes": nodes,
+            "cluster": {
+                "gpu_mem_total": sum(n.total_mem for n in infos),
+                "gpu_mem_used": sum(n.used_mem for n in infos),
+            },
+        },
+        out,
+        indent=1,
+    )
+    out.write("\n")
+
+
 def main(argv=None, kube=None, out=sys.stdout) -> int:
     p = argparse.ArgumentParser(prog="kubectl-inspect-gpushare")
     p.add_argument("-d", "--details", action="store_true",
                    help="per-pod allocation details")
+    p.add_argument("-o", "--output", choices=("table", "json"),
+                   default="table",
+                   help="output format (json: machine-readable dump)")
     p.add_argument("node", nargs="?", default="",
                    help="restrict to one node")
     p.add_argument("--api-url", default=None, help=argparse.SUPPRESS)
@@ -264,7 +314,9 @@ def main(argv=None, kube=None, out=sys.stdout) -> int:
     if not infos:
         print("No shared-GPU nodes found", file=out)
         return 1
-    if args.details:
+    if args.output == "json":
+        display_json(infos, out=out)
+    elif args.details:
         display_details(infos, out=out)
     else:
         display_summary(infos, out=out)

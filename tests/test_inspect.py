@@ -251,3 +251,30 @@ def test_inspect_retry_budget_exhausts():
     with _pytest.raises(ConnectionError):
         inspect_cli._with_retries(always_fails, retries=5, interval=0.001)
     assert calls["n"] == 5
+
+
+def test_json_output_mode():
+    """-o json: machine-readable dump with per-device totals/pods and
+    cluster aggregates."""
+    kube = _cluster()
+    kube.add_pod(make_pod("t1", 4, gpu_idx=0, assigned="true",
+                          phase="Running"))
+    kube.add_pod(make_pod("t2", 6, gpu_idx=1, assigned="true",
+                          phase="Running"))
+    out = io.StringIO()
+    rc = insp.main([], kube=kube, out=out)
+    assert rc == 0  # default table still works
+
+    out = io.StringIO()
+    rc = insp.main(["-o", "json"], kube=kube, out=out)
+    assert rc == 0
+    data = json.loads(out.getvalue())
+    assert data["unit"] in ("GiB", "MiB")
+    assert data["cluster"]["gpu_mem_total"] == 16
+    assert data["cluster"]["gpu_mem_used"] == 10
+    node = data["nodes"][0]
+    assert node["name"] == "node-a"
+    assert node["devices"]["0"]["used"] == 4
+    assert node["devices"]["1"]["used"] == 6
+    pods1 = node["devices"]["1"]["pods"]
+    assert [(p["name"], p["gpu_mem"]) for p in pods1] == [("t2", 6)]
