@@ -289,6 +289,22 @@ void untrack(void* ptr) {
     unreserve(a.size, a.dev);
 }
 
+// fork hygiene: the child must not inherit the parent's pointer→size
+// tracking (freeing an inherited pointer would drain the CHILD's fresh
+// slot into negative territory), and g_sizes_mu must never be forked in
+// a locked state (a parent thread mid-track would deadlock every child
+// allocation forever).
+__attribute__((constructor)) void memguard_fork_handlers() {
+    pthread_atfork(
+        [] { g_sizes_mu.lock(); },
+        [] { g_sizes_mu.unlock(); },
+        [] {
+            g_sizes_mu.unlock();
+            sizes().clear();  // child's own allocs start fresh; bind_slot
+                              // already gives it a zeroed slot
+        });
+}
+
 // clean exit: release the slot immediately (crash/SIGKILL exits are
 // reclaimed lazily by pid_alive sweeps instead)
 __attribute__((destructor)) void memguard_release_slot() {

@@ -323,3 +323,52 @@ os.kill(pid, 9)
         proc.wait(timeout=10)
         if os.path.exists(shm):
             os.unlink(shm)
+
+
+def test_memguard_fork_clears_inherited_tracking():
+    """A forked child inherits the parent's address space but NOT its GPU
+    allocations: freeing an inherited pointer in the child must be a
+    no-op for the child's budget (tracking map cleared at fork), so the
+    child's slot can never go negative."""
+    import os
+    import struct
+    import subprocess
+    import sys
+
+    import gpushare_amd
+
+    lib = os.path.join(
+        os.path.dirname(gpushare_amd.__file__), "libgpushare_memguard.so"
+    )
+    if not os.path.exists(lib):
+        import pytest
+
+        pytest.skip("memguard not built")
+    uid = f"atfork-{os.getpid()}"
+    # exercise via the introspection API: there is no GPU here, so drive
+    # reserve/track indirectly by checking used() stays 0 in the child
+    # and the child owns its own zeroed slot
+    code = f"""
+import ctypes, os, sys, time
+l = ctypes.CDLL({lib!r})
+l.gpushare_memguard_used.restype = ctypes.c_int64
+assert l.gpushare_memguard_used() == 0
+pid = os.fork()
+if pid == 0:
+    # child: fresh slot, zero usage, and allocations tracked by the
+    # parent are invisible (map cleared by the atfork handler)
+    assert l.gpushare_memguard_used() == 0
+    os._exit(0)
+_, status = os.waitpid(pid, 0)
+sys.exit(os.waitstatus_to_exitcode(status))
+"""
+    env = dict(os.environ)
+    env["GPUSHARE_MEM_LIMIT_BYTES"] = str(1 << 30)
+    env["GPUSHARE_POD_UID"] = uid
+    out = subprocess.run(
+        [sys.executable, "-c", code], env=env, capture_output=True, text=True
+    )
+    shm = f"/dev/shm/gpushare.memguard.{uid}.c"
+    if os.path.exists(shm):
+        os.unlink(shm)
+    assert out.returncode == 0, out.stderr[-800:]
