@@ -150,3 +150,26 @@ def test_dockerfile_contract():
     # selftest gates both the build and the container health
     assert dockerfile.count("--selftest") >= 2
     assert "HEALTHCHECK" in dockerfile
+
+
+def test_every_daemon_flag_is_documented():
+    """Doc-drift guard: each daemon flag must appear in
+    docs/operations.md or the daemon docstring's additions list."""
+    import re
+
+    from gpushare_amd.cli import daemon as daemon_mod
+
+    parser_src = open(daemon_mod.__file__.rstrip("c")).read()
+    flags = set(re.findall(r'add_argument\("(--[\w-]+)"', parser_src))
+    ops = open(os.path.join(REPO, "docs", "operations.md")).read()
+    documented = set(re.findall(r"`(--[\w-]+)", ops)) | set(
+        re.findall(r"(--[\w-]+)", parser_src.split('"""')[1])
+    )
+    # short/obvious plumbing flags exempt from the operator-facing table
+    exempt = {
+        "--kubelet-address", "--kubelet-port", "--client-cert",
+        "--client-key", "--token", "--timeout", "--socket-dir",
+        "--mock-spec", "--verbose",
+    }
+    missing = flags - documented - exempt
+    assert not missing, f"undocumented daemon flags: {sorted(missing)}"
